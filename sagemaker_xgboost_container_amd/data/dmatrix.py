@@ -1,0 +1,287 @@
+"""DMatrix — the framework's data container for training and prediction.
+
+Replaces ``xgb.DMatrix`` (the reference's native boundary at
+data_utils.py:309-313,361,384,453). Holds features as either a dense float32
+matrix (missing encoded as NaN) or a scipy CSR matrix, plus per-row label /
+weight / base-margin vectors. Quantization into the on-GPU bin matrix is the
+trainer's job (models/quantize.py) so that one DMatrix can be shared across
+boosters with different ``max_bin``.
+
+Construction accepts:
+  * numpy 2-D arrays / scipy sparse matrices,
+  * file or directory paths, optionally with URI-style parameters:
+        "/path/data?format=csv&label_column=0&delimiter=,&weight_column=1"
+        "/path/data?format=libsvm"
+    (the same URI surface the reference forwards to xgboost,
+    data_utils.py:309-313,361).
+"""
+import os
+from urllib.parse import parse_qs
+
+import numpy as np
+import scipy.sparse as sp
+
+from ..toolkit import exceptions as exc
+
+
+def _is_data_file_name(path, name):
+    if not os.path.isfile(os.path.join(path, name)):
+        return False
+    if name.startswith(".") or name.startswith("_"):
+        return False
+    if ".cache" in name and ("dtrain" in name or "dval" in name):
+        return False
+    return True
+
+
+def _list_data_files(path):
+    if os.path.isfile(path):
+        return [path]
+    files = [os.path.join(path, f) for f in sorted(os.listdir(path)) if _is_data_file_name(path, f)]
+    if not files:
+        raise exc.UserError(f"No data files found under {path}")
+    return files
+
+
+def _parse_csv_files(files, delimiter=",", label_column=0, weight_column=None):
+    import pandas as pd
+
+    frames = [
+        pd.read_csv(f, sep=delimiter, header=None, dtype=np.float32, na_values=[""], skip_blank_lines=True)
+        for f in files
+    ]
+    data = pd.concat(frames, axis=0, ignore_index=True).to_numpy(dtype=np.float32)
+    ncol = data.shape[1]
+    label = None
+    weight = None
+    cols = list(range(ncol))
+    if label_column is not None and ncol > label_column:
+        label = data[:, label_column]
+        cols.remove(label_column)
+    if weight_column is not None and ncol > weight_column:
+        weight = data[:, weight_column]
+        cols.remove(weight_column)
+    features = data[:, cols]
+    return features, label, weight
+
+
+def _parse_libsvm_line(line):
+    parts = line.split()
+    if not parts:
+        return None
+    head = parts[0].split(":")
+    label = float(head[0])
+    weight = float(head[1]) if len(head) == 2 else None
+    indices = []
+    values = []
+    for tok in parts[1:]:
+        idx, _, val = tok.partition(":")
+        indices.append(int(idx))
+        values.append(float(val))
+    return label, weight, indices, values
+
+
+def _parse_libsvm_files(files):
+    """Parse libsvm files (with optional <label>:<weight> extension) to CSR."""
+    labels = []
+    weights = []
+    data = []
+    indices = []
+    indptr = [0]
+    any_weight = False
+    for path in files:
+        with open(path, "r", errors="ignore") as f:
+            for line in f:
+                line = line.strip()
+                if not line or line.startswith("#"):
+                    continue
+                parsed = _parse_libsvm_line(line)
+                if parsed is None:
+                    continue
+                label, weight, idx, val = parsed
+                labels.append(label)
+                if weight is not None:
+                    any_weight = True
+                weights.append(weight if weight is not None else 1.0)
+                indices.extend(idx)
+                data.extend(val)
+                indptr.append(len(indices))
+    ncol = (max(indices) + 1) if indices else 0
+    csr = sp.csr_matrix(
+        (np.asarray(data, dtype=np.float32), np.asarray(indices, dtype=np.int64), np.asarray(indptr, dtype=np.int64)),
+        shape=(len(labels), ncol),
+    )
+    return (
+        csr,
+        np.asarray(labels, dtype=np.float32),
+        np.asarray(weights, dtype=np.float32) if any_weight else None,
+    )
+
+
+def _parse_parquet_files(files):
+    import pyarrow.parquet as pq
+
+    arrays = []
+    for f in files:
+        table = pq.read_table(f)
+        frame = table.to_pandas()
+        arrays.append(np.asarray(frame, dtype=np.float32))
+    data = np.vstack(arrays)
+    return data[:, 1:], data[:, 0], None
+
+
+def _parse_recordio_files(files):
+    from .recordio_protobuf import read_recordio_protobuf
+
+    buf = b"".join(open(f, "rb").read() for f in files)
+    features, labels = read_recordio_protobuf(buf)
+    return features, labels, None
+
+
+def _load_uri(uri):
+    """Load 'path?format=...&k=v' into (features, label, weight)."""
+    path, _, query = uri.partition("?")
+    params = {k: v[0] for k, v in parse_qs(query).items()}
+    fmt = params.get("format")
+    files = _list_data_files(path)
+    if fmt == "csv":
+        return _parse_csv_files(
+            files,
+            delimiter=params.get("delimiter", ","),
+            label_column=int(params["label_column"]) if "label_column" in params else None,
+            weight_column=int(params["weight_column"]) if "weight_column" in params else None,
+        )
+    if fmt == "libsvm" or fmt is None:
+        return _parse_libsvm_files(files)
+    if fmt == "parquet":
+        return _parse_parquet_files(files)
+    if fmt == "recordio-protobuf":
+        return _parse_recordio_files(files)
+    raise exc.UserError(f"Unknown DMatrix format: {fmt}")
+
+
+class DMatrix:
+    """Feature matrix + per-row metadata (label, weight, base margin)."""
+
+    def __init__(
+        self,
+        data,
+        label=None,
+        weight=None,
+        missing=None,
+        feature_names=None,
+        nthread=None,
+        base_margin=None,
+    ):
+        self.missing = np.nan if missing is None else missing
+        self.feature_names = feature_names
+        self._dense = None
+        self._csr = None
+
+        if isinstance(data, str):
+            features, file_label, file_weight = _load_uri(data)
+            if label is None:
+                label = file_label
+            if weight is None:
+                weight = file_weight
+            data = features
+
+        if sp.issparse(data):
+            self._csr = data.tocsr().astype(np.float32)
+        else:
+            arr = np.asarray(data, dtype=np.float32)
+            if arr.ndim == 1:
+                arr = arr.reshape(1, -1)
+            if arr.ndim != 2:
+                raise exc.UserError(f"DMatrix expects 2-D data, got shape {arr.shape}")
+            if not (isinstance(self.missing, float) and np.isnan(self.missing)):
+                arr = arr.copy()
+                arr[arr == self.missing] = np.nan
+            self._dense = np.ascontiguousarray(arr)
+
+        self._label = None if label is None else np.asarray(label, dtype=np.float32).reshape(-1)
+        self._weight = None if weight is None else np.asarray(weight, dtype=np.float32).reshape(-1)
+        self._base_margin = None if base_margin is None else np.asarray(base_margin, dtype=np.float32)
+
+        if self._label is not None and len(self._label) != self.num_row():
+            raise exc.UserError(
+                f"Label length {len(self._label)} does not match number of rows {self.num_row()}"
+            )
+
+    # -- shape ------------------------------------------------------------
+    def num_row(self):
+        return self._dense.shape[0] if self._dense is not None else self._csr.shape[0]
+
+    def num_col(self):
+        return self._dense.shape[1] if self._dense is not None else self._csr.shape[1]
+
+    @property
+    def is_sparse(self):
+        return self._csr is not None
+
+    # -- metadata ---------------------------------------------------------
+    def get_label(self):
+        return self._label if self._label is not None else np.array([], dtype=np.float32)
+
+    def set_label(self, label):
+        self._label = np.asarray(label, dtype=np.float32).reshape(-1)
+
+    def get_weight(self):
+        return self._weight if self._weight is not None else np.array([], dtype=np.float32)
+
+    def set_weight(self, weight):
+        self._weight = None if weight is None else np.asarray(weight, dtype=np.float32).reshape(-1)
+
+    def get_base_margin(self):
+        return self._base_margin
+
+    def set_base_margin(self, margin):
+        self._base_margin = None if margin is None else np.asarray(margin, dtype=np.float32)
+
+    def get_float_info(self, name):
+        if name == "label":
+            return self.get_label()
+        if name == "weight":
+            return self.get_weight()
+        if name == "base_margin":
+            return self._base_margin
+        raise exc.AlgorithmError(f"Unknown float info field: {name}")
+
+    def set_float_info(self, name, value):
+        if name == "label":
+            self.set_label(value)
+        elif name == "weight":
+            self.set_weight(value)
+        elif name == "base_margin":
+            self.set_base_margin(value)
+        else:
+            raise exc.AlgorithmError(f"Unknown float info field: {name}")
+
+    # -- views ------------------------------------------------------------
+    def to_dense(self):
+        """Dense float32 view (NaN = missing). Materializes sparse data."""
+        if self._dense is not None:
+            return self._dense
+        dense = np.asarray(self._csr.todense(), dtype=np.float32)
+        return dense
+
+    def csr(self):
+        if self._csr is not None:
+            return self._csr
+        return sp.csr_matrix(np.nan_to_num(self._dense, nan=0.0))
+
+    def slice(self, rindex):
+        """Row-subset DMatrix (used by k-fold CV, train.py:409-451 parity)."""
+        rindex = np.asarray(rindex, dtype=np.int64)
+        data = self._dense[rindex] if self._dense is not None else self._csr[rindex]
+        out = DMatrix(
+            data,
+            label=self._label[rindex] if self._label is not None else None,
+            weight=self._weight[rindex] if self._weight is not None else None,
+            feature_names=self.feature_names,
+        )
+        return out
+
+    def __repr__(self):
+        kind = "sparse" if self.is_sparse else "dense"
+        return f"DMatrix({self.num_row()}x{self.num_col()}, {kind})"
