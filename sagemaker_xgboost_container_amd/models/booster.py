@@ -1,0 +1,346 @@
+"""Booster: the trained GBT model — prediction + xgboost-format persistence.
+
+Serialization writes the XGBoost Booster JSON schema (learner /
+gradient_booster / trees with split_indices, split_conditions,
+left_children, ...) so models interoperate with the reference container's
+checkpoint/model file protocol (reference train.py:475-486,
+serve_utils.py:171-197, checkpointing.py:372-378). Files saved by upstream
+xgboost in JSON form load here; files saved here load in upstream xgboost.
+Pickled boosters (the serving pickle-first path, serve_utils.py:180-182)
+are supported via plain-numpy state.
+"""
+import json
+import os
+
+import numpy as np
+import torch
+
+from .. import ops
+from .objectives import create_objective
+from .tree import Tree
+
+MODEL_VERSION = [3, 0, 5]  # schema-compatible xgboost version
+
+
+class Booster:
+    def __init__(self, params=None, num_features=0, feature_names=None):
+        self.params = dict(params or {})
+        self.trees = []           # flat list of Tree
+        self.tree_info = []       # class id of each tree (0 for single-output)
+        self.iteration_indptr = [0]
+        self.num_features = num_features
+        self.feature_names = feature_names
+        self.attributes_map = {}
+        self.best_iteration = None
+        self.best_score = None
+        self._objective = None
+        self._predict_cache = None
+
+    # -- basic accessors ---------------------------------------------------
+    @property
+    def objective_name(self):
+        return self.params.get("objective", "reg:squarederror")
+
+    @property
+    def num_class(self):
+        return int(self.params.get("num_class", 0) or 0)
+
+    @property
+    def n_outputs(self):
+        return max(1, self.num_class)
+
+    @property
+    def base_score(self):
+        bs = self.params.get("base_score")
+        return float(bs) if bs is not None else 0.5
+
+    def objective(self):
+        if self._objective is None:
+            self._objective = create_objective(self.objective_name, self.params)
+        return self._objective
+
+    def set_param(self, key, value=None):
+        if isinstance(key, dict):
+            self.params.update(key)
+        else:
+            self.params[key] = value
+        self._objective = None
+
+    def num_boosted_rounds(self):
+        return len(self.iteration_indptr) - 1
+
+    def attributes(self):
+        return dict(self.attributes_map)
+
+    def attr(self, name):
+        return self.attributes_map.get(name)
+
+    def set_attr(self, **kwargs):
+        for k, v in kwargs.items():
+            if v is None:
+                self.attributes_map.pop(k, None)
+            else:
+                self.attributes_map[k] = str(v)
+
+    # -- boosting ----------------------------------------------------------
+    def add_iteration(self, trees, tree_info):
+        """Append one boosting round's trees (n_outputs × num_parallel_tree)."""
+        self.trees.extend(trees)
+        self.tree_info.extend(tree_info)
+        self.iteration_indptr.append(len(self.trees))
+        self._predict_cache = None
+
+    # -- prediction ----------------------------------------------------------
+    def _margin(self, X, iteration_range=None):
+        """Raw margin (n,) or (n, k) for dense float32 tensor X."""
+        obj = self.objective()
+        n = X.shape[0]
+        k = self.n_outputs
+        base = obj.base_margin(self.base_score)
+        device = X.device
+        margin = torch.full((n, k), float(base), dtype=torch.float32, device=device)
+        lo, hi = 0, self.num_boosted_rounds()
+        if iteration_range is not None and iteration_range != (0, 0):
+            lo, hi = iteration_range
+            hi = min(hi, self.num_boosted_rounds()) if hi else self.num_boosted_rounds()
+        backend = ops.backend_for(device)
+        for it in range(lo, hi):
+            for t in range(self.iteration_indptr[it], self.iteration_indptr[it + 1]):
+                margin[:, self.tree_info[t]] += backend.predict_tree(self.trees[t], X)
+        return margin.squeeze(1) if k == 1 else margin
+
+    def predict(
+        self,
+        data,
+        output_margin=False,
+        iteration_range=None,
+        validate_features=True,
+        pred_contribs=False,
+        training=False,
+        ntree_limit=None,
+    ):
+        """Predict for a DMatrix / ndarray. Returns numpy array."""
+        X = self._as_tensor(data, validate_features)
+        if ntree_limit:  # legacy alias: trees -> iterations
+            iteration_range = (0, int(ntree_limit) // max(1, self._trees_per_round()))
+        if pred_contribs:
+            return self._pred_contribs(X)
+        margin = self._margin(X, iteration_range)
+        if output_margin:
+            return margin.cpu().numpy()
+        out = self.objective().transform(margin)
+        return out.cpu().numpy()
+
+    def _trees_per_round(self):
+        if self.num_boosted_rounds() == 0:
+            return self.n_outputs
+        return self.iteration_indptr[1] - self.iteration_indptr[0]
+
+    def _as_tensor(self, data, validate_features=True):
+        from ..data.dmatrix import DMatrix
+
+        if isinstance(data, DMatrix):
+            if validate_features and self.num_features and data.num_col() != self.num_features:
+                raise ValueError(
+                    f"feature_names mismatch: model expects {self.num_features} features, "
+                    f"got {data.num_col()}"
+                )
+            arr = data.to_dense()
+        else:
+            arr = np.asarray(data, dtype=np.float32)
+            if arr.ndim == 1:
+                arr = arr.reshape(1, -1)
+            if validate_features and self.num_features and arr.shape[1] != self.num_features:
+                raise ValueError(
+                    f"feature_names mismatch: model expects {self.num_features} features, "
+                    f"got {arr.shape[1]}"
+                )
+        device = "cuda" if torch.cuda.is_available() and self.params.get("predictor") != "cpu_predictor" else "cpu"
+        return torch.as_tensor(arr, dtype=torch.float32, device=device)
+
+    def _pred_contribs(self, X):
+        """TreeSHAP-style contributions (approximate: Saabas method), host."""
+        n = X.shape[0]
+        f = self.num_features or X.shape[1]
+        Xc = X.cpu().numpy()
+        out = np.zeros((n, f + 1), dtype=np.float32)
+        out[:, -1] = self.objective().base_margin(self.base_score)
+        for tree in self.trees:
+            self._saabas(tree, Xc, out)
+        return out
+
+    @staticmethod
+    def _saabas(tree, X, out):
+        mean_value = tree.value  # base_weights as node expectations
+        for i in range(X.shape[0]):
+            nid = 0
+            while tree.left[nid] >= 0:
+                fv = X[i, tree.feature[nid]]
+                if np.isnan(fv):
+                    nxt = tree.left[nid] if tree.default_left[nid] else tree.right[nid]
+                else:
+                    nxt = tree.left[nid] if fv < tree.threshold[nid] else tree.right[nid]
+                out[i, tree.feature[nid]] += mean_value[nxt] - mean_value[nid]
+                nid = nxt
+            out[i, -1] += mean_value[0]
+
+    # -- serialization -----------------------------------------------------
+    def _tree_to_json(self, tree, tree_id):
+        n = tree.num_nodes
+        leaf = tree.left < 0
+        split_conditions = np.where(leaf, tree.value, tree.threshold).astype(np.float32)
+        return {
+            "base_weights": [float(v) for v in tree.value],
+            "categories": [],
+            "categories_nodes": [],
+            "categories_segments": [],
+            "categories_sizes": [],
+            "default_left": [int(b) for b in tree.default_left],
+            "id": tree_id,
+            "left_children": [int(v) for v in tree.left],
+            "loss_changes": [float(v) for v in tree.gain],
+            "parents": [int(v) if v >= 0 else 2147483647 for v in tree.parent],
+            "right_children": [int(v) for v in tree.right],
+            "split_conditions": [float(v) for v in split_conditions],
+            "split_indices": [int(v) for v in tree.feature],
+            "split_type": [0] * n,
+            "sum_hessian": [float(v) for v in tree.sum_hess],
+            "tree_param": {
+                "num_deleted": "0",
+                "num_feature": str(self.num_features),
+                "num_nodes": str(n),
+                "size_leaf_vector": "1",
+            },
+        }
+
+    @staticmethod
+    def _tree_from_json(obj):
+        left = np.asarray(obj["left_children"], dtype=np.int32)
+        cond = np.asarray(obj["split_conditions"], dtype=np.float32)
+        leaf = left < 0
+        tree = Tree.from_arrays(
+            {
+                "left": left,
+                "right": np.asarray(obj["right_children"], dtype=np.int32),
+                "parent": [(-1 if p == 2147483647 else p) for p in obj.get("parents", [2147483647] * len(left))],
+                "feature": np.asarray(obj["split_indices"], dtype=np.int32),
+                "threshold": np.where(leaf, 0.0, cond).astype(np.float32),
+                "default_left": np.asarray(obj["default_left"], dtype=bool),
+                "value": np.where(leaf, cond, np.asarray(obj.get("base_weights", cond), dtype=np.float32)),
+                "gain": obj.get("loss_changes", np.zeros(len(left))),
+                "sum_hessian": obj.get("sum_hessian", np.zeros(len(left))),
+            }
+        )
+        tree.sum_hess = np.asarray(obj.get("sum_hessian", np.zeros(len(left))), dtype=np.float32)
+        return tree
+
+    def _objective_json(self):
+        name = self.objective_name
+        obj = {"name": name}
+        if name.startswith("binary:") or name == "reg:logistic":
+            obj["reg_loss_param"] = {"scale_pos_weight": str(self.params.get("scale_pos_weight", 1.0))}
+        elif name.startswith("multi:"):
+            obj["softmax_multiclass_param"] = {"num_class": str(self.num_class)}
+        elif name == "count:poisson":
+            obj["poisson_regression_param"] = {"max_delta_step": str(self.params.get("max_delta_step", 0.7))}
+        elif name == "reg:tweedie":
+            obj["tweedie_regression_param"] = {
+                "tweedie_variance_power": str(self.params.get("tweedie_variance_power", 1.5))
+            }
+        return obj
+
+    def save_json(self):
+        model = {
+            "learner": {
+                "attributes": dict(self.attributes_map),
+                "feature_names": self.feature_names or [],
+                "feature_types": [],
+                "gradient_booster": {
+                    "model": {
+                        "gbtree_model_param": {
+                            "num_trees": str(len(self.trees)),
+                            "num_parallel_tree": str(self.params.get("num_parallel_tree", 1)),
+                        },
+                        "iteration_indptr": list(self.iteration_indptr),
+                        "tree_info": list(self.tree_info),
+                        "trees": [self._tree_to_json(t, i) for i, t in enumerate(self.trees)],
+                    },
+                    "name": "gbtree",
+                },
+                "learner_model_param": {
+                    "base_score": repr(self.base_score),
+                    "boost_from_average": "1",
+                    "num_class": str(self.num_class),
+                    "num_feature": str(self.num_features),
+                    "num_target": "1",
+                },
+                "objective": self._objective_json(),
+            },
+            "version": MODEL_VERSION,
+        }
+        return model
+
+    def save_model(self, path):
+        with open(str(path) + ".tmp", "w") as f:
+            json.dump(self.save_json(), f)
+        os.replace(str(path) + ".tmp", str(path))
+
+    def load_json(self, model):
+        learner = model["learner"]
+        gb = learner["gradient_booster"]["model"]
+        lmp = learner["learner_model_param"]
+        self.params["objective"] = learner["objective"]["name"]
+        if int(lmp.get("num_class", "0") or 0) > 0:
+            self.params["num_class"] = int(lmp["num_class"])
+        self.params["base_score"] = float(lmp.get("base_score", 0.5))
+        self.num_features = int(lmp.get("num_feature", 0))
+        self.feature_names = learner.get("feature_names") or None
+        self.attributes_map = dict(learner.get("attributes", {}))
+        self.trees = [self._tree_from_json(t) for t in gb["trees"]]
+        self.tree_info = list(gb.get("tree_info", [0] * len(self.trees)))
+        indptr = gb.get("iteration_indptr")
+        if indptr:
+            self.iteration_indptr = list(indptr)
+        else:
+            per_round = max(1, self.n_outputs)
+            self.iteration_indptr = list(range(0, len(self.trees) + 1, per_round))
+        self._objective = None
+        self._predict_cache = None
+        return self
+
+    def load_model(self, path):
+        with open(path, "rb") as f:
+            head = f.read(1)
+        if head not in (b"{",):
+            raise ValueError(f"Unsupported model format in {path} (expected JSON Booster)")
+        with open(path, "r") as f:
+            return self.load_json(json.load(f))
+
+    def save_config(self):
+        return json.dumps(
+            {
+                "learner": {
+                    "learner_train_param": {"objective": self.objective_name},
+                    "learner_model_param": {
+                        "num_class": str(self.num_class),
+                        "base_score": repr(self.base_score),
+                        "num_feature": str(self.num_features),
+                    },
+                }
+            }
+        )
+
+    # -- pickling ----------------------------------------------------------
+    def __getstate__(self):
+        state = dict(self.__dict__)
+        state["_objective"] = None
+        state["_predict_cache"] = None
+        return state
+
+    def __setstate__(self, state):
+        self.__dict__.update(state)
+
+    def copy(self):
+        import copy as _copy
+
+        return _copy.deepcopy(self)

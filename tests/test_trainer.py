@@ -1,0 +1,348 @@
+"""Core GBT engine tests: objectives, growing, training loop, serialization.
+
+CPU (torch reference backend). These define the numerics the HIP kernels are
+validated against in tests/test_gpu_kernels.py.
+"""
+import json
+
+import numpy as np
+import pytest
+import torch
+
+from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+from sagemaker_xgboost_container_amd.models import trainer
+from sagemaker_xgboost_container_amd.models.booster import Booster
+from sagemaker_xgboost_container_amd.models.callback_api import EarlyStopping
+from sagemaker_xgboost_container_amd.ops.quantize import quantize
+
+
+def _binary_data(n=2000, f=10, seed=0):
+    rng = np.random.default_rng(seed)
+    X = rng.normal(size=(n, f)).astype(np.float32)
+    logit = X[:, 0] * 2 - X[:, 1] + 0.5 * X[:, 2] * X[:, 3]
+    y = (logit + rng.normal(scale=0.5, size=n) > 0).astype(np.float32)
+    return X, y
+
+
+class TestQuantize:
+    def test_bins_and_cuts(self):
+        X = torch.tensor([[0.0], [1.0], [2.0], [3.0]], dtype=torch.float32)
+        qm = quantize(X, max_bin=256)
+        assert int(qm.nbins[0]) == 4
+        assert qm.bins[:, 0].tolist() == [0, 1, 2, 3]
+
+    def test_quantile_binning_large(self):
+        rng = np.random.default_rng(0)
+        X = torch.tensor(rng.normal(size=(10000, 2)).astype(np.float32))
+        qm = quantize(X, max_bin=64)
+        assert int(qm.nbins.max()) <= 64
+        # roughly uniform occupancy
+        counts = torch.bincount(qm.bins[:, 0].long(), minlength=int(qm.nbins[0]))
+        assert counts.min() > 30
+
+    def test_missing(self):
+        X = torch.tensor([[0.0], [float("nan")], [2.0]], dtype=torch.float32)
+        qm = quantize(X, max_bin=8)
+        assert qm.has_missing
+        assert int(qm.bins[1, 0]) == qm.stride - 1
+
+    def test_eval_binning_with_train_cuts(self):
+        Xtr = torch.tensor(np.random.default_rng(0).normal(size=(500, 3)).astype(np.float32))
+        qm = quantize(Xtr, max_bin=32)
+        Xev = Xtr[:100] + 0.01
+        qm2 = quantize(Xev, max_bin=32, cuts=qm.cuts, cut_ptr=qm.cut_ptr, nbins=qm.nbins)
+        assert qm2.stride <= qm.stride + 1
+
+
+class TestTraining:
+    def test_binary_logistic_learns(self):
+        X, y = _binary_data()
+        dtrain = DMatrix(X[:1500], label=y[:1500])
+        dval = DMatrix(X[1500:], label=y[1500:])
+        res = {}
+        bst = trainer.train(
+            {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3},
+            dtrain,
+            num_boost_round=20,
+            evals=[(dtrain, "train"), (dval, "validation")],
+            evals_result=res,
+            verbose_eval=False,
+        )
+        assert res["train"]["logloss"][-1] < 0.25
+        pred = bst.predict(dval)
+        assert ((pred > 0.5) == y[1500:]).mean() > 0.85
+
+    def test_regression_learns(self):
+        rng = np.random.default_rng(1)
+        X = rng.normal(size=(1000, 5)).astype(np.float32)
+        y = (3 * X[:, 0] - 2 * X[:, 1] ** 2 + rng.normal(scale=0.1, size=1000)).astype(np.float32)
+        dtrain = DMatrix(X, label=y)
+        res = {}
+        trainer.train(
+            {"objective": "reg:squarederror", "max_depth": 5, "eta": 0.3},
+            dtrain,
+            num_boost_round=30,
+            evals=[(dtrain, "train")],
+            evals_result=res,
+            verbose_eval=False,
+        )
+        assert res["train"]["rmse"][-1] < 0.8 * res["train"]["rmse"][0]
+
+    def test_multiclass_softprob(self):
+        rng = np.random.default_rng(2)
+        n = 1500
+        X = rng.normal(size=(n, 6)).astype(np.float32)
+        y = (X[:, 0] > 0.5).astype(int) + (X[:, 1] > 0).astype(int)  # 3 classes
+        dtrain = DMatrix(X, label=y.astype(np.float32))
+        res = {}
+        bst = trainer.train(
+            {"objective": "multi:softprob", "num_class": 3, "max_depth": 4, "eta": 0.4},
+            dtrain,
+            num_boost_round=10,
+            evals=[(dtrain, "train")],
+            evals_result=res,
+            verbose_eval=False,
+        )
+        assert res["train"]["mlogloss"][-1] < 0.4
+        probs = bst.predict(dtrain)
+        assert probs.shape == (n, 3)
+        np.testing.assert_allclose(probs.sum(axis=1), 1.0, atol=1e-5)
+        assert (probs.argmax(axis=1) == y).mean() > 0.85
+
+    def test_multiclass_softmax_predicts_labels(self):
+        rng = np.random.default_rng(3)
+        X = rng.normal(size=(600, 4)).astype(np.float32)
+        y = (X[:, 0] > 0).astype(np.float32)
+        dtrain = DMatrix(X, label=y)
+        bst = trainer.train(
+            {"objective": "multi:softmax", "num_class": 2, "max_depth": 3},
+            dtrain,
+            num_boost_round=5,
+            verbose_eval=False,
+        )
+        pred = bst.predict(dtrain)
+        assert set(np.unique(pred)).issubset({0.0, 1.0})
+
+    def test_missing_values(self):
+        X, y = _binary_data(1000)
+        X[::7, 0] = np.nan
+        X[::11, 3] = np.nan
+        dtrain = DMatrix(X, label=y)
+        res = {}
+        trainer.train(
+            {"objective": "binary:logistic", "max_depth": 4},
+            dtrain,
+            num_boost_round=10,
+            evals=[(dtrain, "train")],
+            evals_result=res,
+            verbose_eval=False,
+        )
+        assert res["train"]["logloss"][-1] < res["train"]["logloss"][0]
+
+    def test_weighted_training(self):
+        X, y = _binary_data(800)
+        w = np.where(y == 1, 5.0, 1.0).astype(np.float32)
+        dtrain = DMatrix(X, label=y, weight=w)
+        bst = trainer.train(
+            {"objective": "binary:logistic", "max_depth": 3}, dtrain, num_boost_round=5, verbose_eval=False
+        )
+        assert bst.num_boosted_rounds() == 5
+
+    def test_subsample_colsample(self):
+        X, y = _binary_data(800)
+        dtrain = DMatrix(X, label=y)
+        bst = trainer.train(
+            {
+                "objective": "binary:logistic",
+                "max_depth": 4,
+                "subsample": 0.7,
+                "colsample_bytree": 0.7,
+                "colsample_bylevel": 0.8,
+                "colsample_bynode": 0.8,
+                "seed": 7,
+            },
+            dtrain,
+            num_boost_round=5,
+            verbose_eval=False,
+        )
+        assert bst.num_boosted_rounds() == 5
+
+    def test_lossguide(self):
+        X, y = _binary_data(1000)
+        dtrain = DMatrix(X, label=y)
+        res = {}
+        bst = trainer.train(
+            {"objective": "binary:logistic", "grow_policy": "lossguide", "max_leaves": 15, "max_depth": 0},
+            dtrain,
+            num_boost_round=5,
+            evals=[(dtrain, "train")],
+            evals_result=res,
+            verbose_eval=False,
+        )
+        for t in bst.trees:
+            assert t.num_leaves <= 15
+        assert res["train"]["logloss"][-1] < res["train"]["logloss"][0]
+
+    def test_early_stopping(self):
+        X, y = _binary_data(600)
+        noise_val_X = np.random.default_rng(9).normal(size=(200, 10)).astype(np.float32)
+        noise_val_y = np.random.default_rng(10).integers(0, 2, 200).astype(np.float32)
+        dtrain = DMatrix(X, label=y)
+        dval = DMatrix(noise_val_X, label=noise_val_y)
+        bst = trainer.train(
+            {"objective": "binary:logistic", "max_depth": 6, "eta": 0.5},
+            dtrain,
+            num_boost_round=100,
+            evals=[(dval, "validation")],
+            early_stopping_rounds=5,
+            verbose_eval=False,
+        )
+        assert bst.num_boosted_rounds() < 100
+        assert bst.best_iteration is not None
+
+    def test_early_stopping_save_best(self):
+        X, y = _binary_data(600)
+        dtrain = DMatrix(X, label=y)
+        dval = DMatrix(X[:200] + 3.0, label=1 - y[:200])
+        bst = trainer.train(
+            {"objective": "binary:logistic", "max_depth": 6, "eta": 0.6},
+            dtrain,
+            num_boost_round=50,
+            evals=[(dval, "validation")],
+            callbacks=[EarlyStopping(rounds=3, save_best=True)],
+            verbose_eval=False,
+        )
+        assert bst.num_boosted_rounds() == bst.best_iteration + 1
+
+    def test_warm_start(self):
+        X, y = _binary_data(600)
+        dtrain = DMatrix(X, label=y)
+        params = {"objective": "binary:logistic", "max_depth": 3}
+        b1 = trainer.train(params, dtrain, num_boost_round=3, verbose_eval=False)
+        b2 = trainer.train(params, dtrain, num_boost_round=2, xgb_model=b1, verbose_eval=False)
+        assert b2.num_boosted_rounds() == 5
+
+    def test_base_score(self):
+        X, y = _binary_data(400)
+        dtrain = DMatrix(X, label=y)
+        bst = trainer.train(
+            {"objective": "binary:logistic", "base_score": "0.2", "max_depth": 2},
+            dtrain,
+            num_boost_round=1,
+            verbose_eval=False,
+        )
+        assert bst.base_score == 0.2
+
+    def test_num_parallel_tree(self):
+        X, y = _binary_data(400)
+        dtrain = DMatrix(X, label=y)
+        bst = trainer.train(
+            {"objective": "binary:logistic", "max_depth": 3, "num_parallel_tree": 2, "subsample": 0.6},
+            dtrain,
+            num_boost_round=3,
+            verbose_eval=False,
+        )
+        assert len(bst.trees) == 6
+
+    def test_objective_label_validation(self):
+        X, _ = _binary_data(100)
+        dtrain = DMatrix(X, label=np.full(100, 2.0, dtype=np.float32))
+        with pytest.raises(ValueError, match="label must be in \\[0,1\\]"):
+            trainer.train({"objective": "binary:logistic"}, dtrain, num_boost_round=1, verbose_eval=False)
+
+
+class TestSerialization:
+    def _trained(self):
+        X, y = _binary_data(500)
+        dtrain = DMatrix(X, label=y)
+        bst = trainer.train(
+            {"objective": "binary:logistic", "max_depth": 4}, dtrain, num_boost_round=5, verbose_eval=False
+        )
+        return bst, X
+
+    def test_json_round_trip(self, tmp_path):
+        bst, X = self._trained()
+        path = tmp_path / "xgboost-model"
+        bst.save_model(path)
+        loaded = Booster()
+        loaded.load_model(path)
+        np.testing.assert_allclose(bst.predict(X), loaded.predict(X), rtol=1e-6)
+        assert loaded.objective_name == "binary:logistic"
+
+    def test_json_schema_fields(self, tmp_path):
+        bst, _ = self._trained()
+        path = tmp_path / "m.json"
+        bst.save_model(path)
+        obj = json.load(open(path))
+        assert obj["version"] == [3, 0, 5]
+        learner = obj["learner"]
+        assert learner["objective"]["name"] == "binary:logistic"
+        trees = learner["gradient_booster"]["model"]["trees"]
+        assert len(trees) == 5
+        t0 = trees[0]
+        for key in ("left_children", "right_children", "split_indices", "split_conditions",
+                    "default_left", "base_weights", "loss_changes", "sum_hessian", "tree_param"):
+            assert key in t0
+        assert t0["tree_param"]["num_nodes"] == str(len(t0["left_children"]))
+
+    def test_pickle_round_trip(self):
+        import pickle
+
+        bst, X = self._trained()
+        clone = pickle.loads(pickle.dumps(bst))
+        np.testing.assert_allclose(bst.predict(X), clone.predict(X), rtol=1e-6)
+
+    def test_predict_margin_and_iteration_range(self):
+        bst, X = self._trained()
+        m = bst.predict(X, output_margin=True)
+        p = bst.predict(X)
+        np.testing.assert_allclose(p, 1 / (1 + np.exp(-m)), rtol=1e-5)
+        p2 = bst.predict(X, iteration_range=(0, 2))
+        assert not np.allclose(p, p2)
+
+    def test_feature_mismatch(self):
+        bst, X = self._trained()
+        with pytest.raises(ValueError, match="feature_names mismatch"):
+            bst.predict(X[:, :5])
+
+    def test_pred_contribs_sums_to_margin(self):
+        bst, X = self._trained()
+        contribs = bst.predict(DMatrix(X[:50], label=np.zeros(50)), pred_contribs=True)
+        margin = bst.predict(X[:50], output_margin=True)
+        np.testing.assert_allclose(contribs.sum(axis=1), margin, rtol=1e-4, atol=1e-4)
+
+
+class TestObjectiveZoo:
+    @pytest.mark.parametrize(
+        "objective,extra",
+        [
+            ("reg:squarederror", {}),
+            ("reg:linear", {}),
+            ("reg:logistic", {}),
+            ("binary:logitraw", {}),
+            ("binary:hinge", {}),
+            ("count:poisson", {}),
+            ("reg:gamma", {}),
+            ("reg:tweedie", {}),
+            ("reg:pseudohubererror", {}),
+            ("reg:absoluteerror", {}),
+            ("reg:squaredlogerror", {}),
+        ],
+    )
+    def test_objective_trains(self, objective, extra):
+        rng = np.random.default_rng(0)
+        X = rng.normal(size=(500, 5)).astype(np.float32)
+        raw = X[:, 0] - 0.5 * X[:, 1]
+        if objective.startswith(("count:", "reg:gamma", "reg:tweedie")):
+            y = np.exp(raw * 0.3).astype(np.float32) + 0.1
+        elif objective.startswith("binary") or objective == "reg:logistic":
+            y = (raw > 0).astype(np.float32)
+        elif objective == "reg:squaredlogerror":
+            y = np.abs(raw).astype(np.float32)
+        else:
+            y = raw.astype(np.float32)
+        params = {"objective": objective, "max_depth": 3, **extra}
+        dtrain = DMatrix(X, label=y)
+        bst = trainer.train(params, dtrain, num_boost_round=3, verbose_eval=False)
+        pred = bst.predict(dtrain)
+        assert np.isfinite(pred).all()
