@@ -1,21 +1,28 @@
 """Histogram-based tree grower (the `hist` / `gpu_hist` updater).
 
 Replaces xgboost's grow_quantile_histmaker / updater_gpu_hist (the native
-hot loop behind xgb.train). Per level:
+hot loop behind xgb.train). The design is level-synchronous and
+segment-based — all per-level device work is batched so a level costs a
+fixed small number of kernel launches and exactly ONE device→host sync (the
+partition left-counts readback):
 
-  1. build (grad, hess) histograms for the new frontier — only the smaller
-     child of each sibling pair is built, the larger is derived by the
-     subtraction trick from the cached parent histogram;
-  2. (distributed) allreduce the level's histograms across ranks — one
-     fused buffer per level over RCCL/xGMI;
-  3. split-gain scan over all (feature, bin) candidates incl. both missing
-     directions;
-  4. partition each split node's rows into child row sets.
+  1. histogram build for the new frontier — only the smaller child (by
+     global hessian) of each sibling pair is built; the larger is derived by
+     the subtraction trick from the cached parent histogram;
+  2. (distributed) ONE fused allreduce of the level's histogram accumulator
+     across ranks over RCCL/xGMI (int64 fixed-point on GPU: the sum is
+     bit-deterministic and rank-order independent);
+  3. batched split-gain scan over all (feature, bin) candidates including
+     both missing directions;
+  4. batched partition of all split nodes' row segments into the
+     opposite-parity row buffer (two-ended compaction in the same range).
 
-Backend ops (build_histogram / find_splits / partition_rows) come from
-ops.backend_for(device): CDNA4 HIP kernels on MI355X, torch reference on CPU.
-Supports depthwise and lossguide grow policies, row subsampling and
-colsample_by{tree,level,node}.
+Row sets live in two ping-pong int32 buffers; a node's segment is
+(parity, start, end) and children occupy the parent's range in the other
+buffer, so leaves finalized at different depths coexist without copies.
+
+Backend ops come from ops.backend_for(device): CDNA4 HIP kernels on MI355X,
+torch reference on CPU.
 """
 import heapq
 
@@ -45,16 +52,34 @@ class GrowParams:
             self.max_depth = 6
 
 
+class _Node:
+    __slots__ = ("nid", "parity", "start", "end", "g", "h", "depth")
+
+    def __init__(self, nid, parity, start, end, g, h, depth):
+        self.nid = nid
+        self.parity = parity
+        self.start = start
+        self.end = end
+        self.g = g
+        self.h = h
+        self.depth = depth
+
+    @property
+    def seg(self):
+        return (self.start, self.end)
+
+
 class HistGrower:
     def __init__(self, qm, params, comm=None, generator=None):
         self.qm = qm
         self.p = GrowParams(params)
-        self.comm = comm  # optional: object with allreduce_(tensor)
+        self.comm = comm  # optional: allreduce_(tensor) / allreduce_max_(tensor) / rank
         self.backend = ops.backend_for(qm.device)
         self.device = qm.device
         self.generator = generator
+        self._bufs = None
 
-    # -- weight / gain math (host scalars; double precision) --------------
+    # -- weight / gain math (host scalars; double precision) ---------------
     def _weight(self, g, h):
         a = abs(g) - self.p.reg_alpha
         if a < 0:
@@ -63,12 +88,6 @@ class HistGrower:
         if self.p.max_delta_step > 0:
             w = max(-self.p.max_delta_step, min(self.p.max_delta_step, w))
         return w
-
-    def _colsample_mask(self, base_mask):
-        f = self.qm.num_col
-        if base_mask is None:
-            base_mask = torch.ones(f, dtype=torch.bool, device=self.device)
-        return base_mask
 
     def _sample_features(self, frac, prev_mask):
         f = self.qm.num_col
@@ -86,88 +105,101 @@ class HistGrower:
             self.comm.allreduce_(tensor)
         return tensor
 
-    def _build_hists(self, gh, build_jobs, derived_jobs, node_hist):
-        """build_jobs: [(nid, rows)]; derived: [(nid, parent_nid, sibling_nid)].
-
-        Histograms for build_jobs are summed across ranks (one fused
-        allreduce); derived nodes use subtraction (already-global values).
-        """
-        slots = self.qm.total_slots
-        if build_jobs:
-            stack = torch.zeros((len(build_jobs), slots, 2), dtype=torch.float32, device=self.device)
-            for i, (nid, rows) in enumerate(build_jobs):
-                self.backend.build_histogram(self.qm, gh, rows, out=stack[i])
-            self._allreduce(stack)
-            for i, (nid, _rows) in enumerate(build_jobs):
-                node_hist[nid] = stack[i]
-        for nid, parent_nid, sibling_nid in derived_jobs:
+    def _build_level_hists(self, gh, scale, build_nodes, derived, node_hist):
+        """build_nodes: [_Node] (all same parity); derived: [(nid, parent, sib)]."""
+        if build_nodes:
+            rowbuf = self._bufs[build_nodes[0].parity]
+            acc = self.backend.build_histograms(
+                self.qm, gh, rowbuf, [n.seg for n in build_nodes], scale
+            )
+            self._allreduce(acc)
+            hist = self.backend.hist_to_float(acc, scale)
+            for i, node in enumerate(build_nodes):
+                node_hist[node.nid] = hist[i]
+        for nid, parent_nid, sibling_nid in derived:
             node_hist[nid] = node_hist[parent_nid] - node_hist[sibling_nid]
 
-    def grow(self, gh, rows=None):
+    def grow(self, gh, sample_rows=None):
         """Grow one tree from (n, 2) float32 gradients.
 
-        Returns (tree, leaf_segments) where leaf_segments is
-        {leaf_id: row_idx_tensor} of this rank's rows per leaf.
+        Returns (tree, leaf_jobs) where leaf_jobs is a list of
+        (parity, start, end, leaf_value) covering this rank's rows; pass to
+        backend.update_margins together with self.bufs.
         """
         qm = self.qm
         p = self.p
         n = qm.num_row
-        if rows is None:
+        if sample_rows is not None:
+            rows = sample_rows
+        elif p.subsample < 1.0:
+            keep = torch.rand(n, device=self.device, generator=self.generator) < p.subsample
+            rows = keep.nonzero(as_tuple=True)[0].to(torch.int32)
+        else:
             rows = torch.arange(n, dtype=torch.int32, device=self.device)
-        if p.subsample < 1.0:
-            keep = torch.rand(rows.numel(), device=self.device, generator=self.generator) < p.subsample
-            rows = rows[keep]
+
+        cap = rows.numel()
+        if self._bufs is None or self._bufs[0].numel() < cap:
+            self._bufs = (
+                torch.empty(max(cap, n), dtype=torch.int32, device=self.device),
+                torch.empty(max(cap, n), dtype=torch.int32, device=self.device),
+            )
+        self._bufs[0][:cap] = rows
+
+        scale = self.backend.compute_scale(gh, comm=self.comm)
 
         tree = Tree()
         root_sum = gh.index_select(0, rows.long()).to(torch.float64).sum(0)
         root_sum = self._allreduce(root_sum)
         G, H = float(root_sum[0]), float(root_sum[1])
         root = tree.add_node(parent=-1, value=self._weight(G, H) * p.eta, sum_hess=H)
+        root_node = _Node(root, 0, 0, cap, G, H, 0)
 
         tree_mask = self._sample_features(p.colsample_bytree, None)
 
-        node_rows = {root: rows}
-        node_hist = {}
-        node_sum = {root: (G, H)}
-
         if p.grow_policy == "lossguide":
-            leaf_segments = self._grow_lossguide(tree, gh, node_rows, node_hist, node_sum, tree_mask)
+            leaf_jobs = self._grow_lossguide(tree, gh, scale, root_node, tree_mask)
         else:
-            leaf_segments = self._grow_depthwise(tree, gh, node_rows, node_hist, node_sum, tree_mask)
-        return tree, leaf_segments
+            leaf_jobs = self._grow_depthwise(tree, gh, scale, root_node, tree_mask)
+        return tree, leaf_jobs
 
-    # -- depthwise ---------------------------------------------------------
-    def _grow_depthwise(self, tree, gh, node_rows, node_hist, node_sum, tree_mask):
+    @property
+    def bufs(self):
+        return self._bufs
+
+    # -- depthwise ----------------------------------------------------------
+    def _grow_depthwise(self, tree, gh, scale, root_node, tree_mask):
         p = self.p
-        frontier = [0]  # node ids whose hist must be considered for splitting
-        depth = 0
+        node_hist = {}
+        finished = []  # _Node leaves
+        frontier = [root_node]
         n_leaves = 1
+        depth = 0
         while frontier and (p.max_depth == 0 or depth < p.max_depth):
             if p.max_leaves and n_leaves >= p.max_leaves:
                 break
             level_mask = self._sample_features(p.colsample_bylevel, tree_mask)
 
-            # 1. histograms: root is built; children pairs use subtraction
-            build_jobs, derived_jobs = [], []
-            for nid in frontier:
-                parent = int(tree.parent[nid])
+            # 1. histograms (smaller-by-global-hessian child built, sibling derived)
+            build_nodes, derived = [], []
+            for node in frontier:
+                parent = int(tree.parent[node.nid])
                 if parent < 0:
-                    build_jobs.append((nid, node_rows[nid]))
+                    build_nodes.append(node)
                 else:
-                    sibling = int(tree.right[parent]) if int(tree.left[parent]) == nid else int(tree.left[parent])
-                    # build the smaller child, derive the larger. Compare by
-                    # GLOBAL hessian sum (not local row count) so every rank
-                    # makes the same choice and allreduce buffers line up.
-                    if node_sum[nid][1] <= node_sum[sibling][1]:
-                        build_jobs.append((nid, node_rows[nid]))
+                    sib_nid = (
+                        int(tree.right[parent]) if int(tree.left[parent]) == node.nid else int(tree.left[parent])
+                    )
+                    sibling = next(m for m in frontier if m.nid == sib_nid)
+                    if (node.h, node.nid) <= (sibling.h, sib_nid):
+                        build_nodes.append(node)
                     else:
-                        derived_jobs.append((nid, parent, sibling))
-            self._build_hists(gh, build_jobs, derived_jobs, node_hist)
+                        derived.append((node.nid, parent, sib_nid))
+            self._build_level_hists(gh, scale, build_nodes, derived, node_hist)
 
             # 2. batched split search
-            hists = torch.stack([node_hist[nid] for nid in frontier])
+            hists = torch.stack([node_hist[node.nid] for node in frontier])
             parent_sums = torch.tensor(
-                [node_sum[nid] for nid in frontier], dtype=torch.float32, device=self.device
+                [(node.g, node.h) for node in frontier], dtype=torch.float32, device=self.device
             )
             node_feature_mask = self._sample_features(p.colsample_bynode, level_mask)
             splits = self.backend.find_splits(
@@ -187,65 +219,76 @@ class HistGrower:
             lgs = splits["left_g"].cpu().numpy()
             lhs = splits["left_h"].cpu().numpy()
 
-            next_frontier = []
-            for i, nid in enumerate(frontier):
-                if gains[i] <= 0.0:
-                    self._free_parent(nid, tree, node_hist)
-                    continue
-                if p.max_leaves and n_leaves >= p.max_leaves:
-                    self._free_parent(nid, tree, node_hist)
-                    continue
-                lid, rid = self._apply_split(
-                    tree, nid, int(feats[i]), int(bins[i]), bool(dls[i]),
-                    float(gains[i]), float(lgs[i]), float(lhs[i]), node_sum,
-                )
-                left_rows, right_rows = self.backend.partition_rows(
-                    self.qm, node_rows[nid], int(feats[i]), int(bins[i]), bool(dls[i])
-                )
-                node_rows[lid] = left_rows
-                node_rows[rid] = right_rows
-                del node_rows[nid]
-                n_leaves += 1
-                next_frontier += [lid, rid]
+            # 3. decide splits (leaf cap) and batch-partition
+            to_split = []
+            for i, node in enumerate(frontier):
+                if gains[i] <= 0.0 or (p.max_leaves and n_leaves >= p.max_leaves):
+                    finished.append(node)
+                else:
+                    to_split.append((i, node))
+                    n_leaves += 1
 
-            # free grandparent hists (parents of this frontier are done)
-            for nid in frontier:
-                parent = int(tree.parent[nid])
+            next_frontier = []
+            if to_split:
+                parity = to_split[0][1].parity
+                counts = self.backend.partition_level(
+                    self.qm,
+                    self._bufs[parity],
+                    self._bufs[1 - parity],
+                    [node.seg for _i, node in to_split],
+                    [int(feats[i]) for i, _n in to_split],
+                    [int(bins[i]) for i, _n in to_split],
+                    [bool(dls[i]) for i, _n in to_split],
+                )
+                for (i, node), left_count in zip(to_split, counts):
+                    lid, rid = self._apply_split(
+                        tree, node, int(feats[i]), int(bins[i]), bool(dls[i]),
+                        float(gains[i]), float(lgs[i]), float(lhs[i]),
+                    )
+                    mid = node.start + left_count
+                    next_frontier.append(
+                        _Node(lid, 1 - parity, node.start, mid, float(lgs[i]), float(lhs[i]), depth + 1)
+                    )
+                    next_frontier.append(
+                        _Node(
+                            rid, 1 - parity, mid, node.end,
+                            node.g - float(lgs[i]), node.h - float(lhs[i]), depth + 1,
+                        )
+                    )
+
+            # free hists no longer needed (grandparents of the next frontier)
+            for node in frontier:
+                parent = int(tree.parent[node.nid])
                 if parent >= 0:
                     node_hist.pop(parent, None)
             frontier = next_frontier
             depth += 1
 
-        for nid in frontier:
-            self._free_parent(nid, tree, node_hist)
+        finished.extend(frontier)
         node_hist.clear()
-        return node_rows
+        return [(n.parity, n.start, n.end, float(tree.value[n.nid])) for n in finished]
 
-    def _free_parent(self, nid, tree, node_hist):
-        pass  # hist cleanup handled at level end
-
-    # -- lossguide ---------------------------------------------------------
-    def _grow_lossguide(self, tree, gh, node_rows, node_hist, node_sum, tree_mask):
+    # -- lossguide ----------------------------------------------------------
+    def _grow_lossguide(self, tree, gh, scale, root_node, tree_mask):
         p = self.p
         max_leaves = p.max_leaves if p.max_leaves else 2 ** max(p.max_depth, 1)
+        node_hist = {}
+        nodes = {root_node.nid: root_node}
         counter = 0
-        heap = []  # (-gain, counter, nid, split dict)
+        heap = []
 
-        def evaluate(nid, depth):
+        def evaluate(node, sibling=None):
             nonlocal counter
-            parent = int(tree.parent[nid])
-            if parent < 0:
-                self._build_hists(gh, [(nid, node_rows[nid])], [], node_hist)
-            else:
-                sibling = int(tree.right[parent]) if int(tree.left[parent]) == nid else int(tree.left[parent])
-                if sibling in node_hist and nid not in node_hist:
-                    self._build_hists(gh, [], [(nid, parent, sibling)], node_hist)
-                elif nid not in node_hist:
-                    self._build_hists(gh, [(nid, node_rows[nid])], [], node_hist)
+            parent = int(tree.parent[node.nid])
+            if node.nid not in node_hist:
+                if parent >= 0 and sibling is not None and sibling.nid in node_hist:
+                    self._build_level_hists(gh, scale, [], [(node.nid, parent, sibling.nid)], node_hist)
+                else:
+                    self._build_level_hists(gh, scale, [node], [], node_hist)
             mask = self._sample_features(p.colsample_bynode, tree_mask)
             s = self.backend.find_splits(
-                node_hist[nid].unsqueeze(0),
-                torch.tensor([node_sum[nid]], dtype=torch.float32, device=self.device),
+                node_hist[node.nid].unsqueeze(0),
+                torch.tensor([(node.g, node.h)], dtype=torch.float32, device=self.device),
                 self.qm,
                 reg_lambda=p.reg_lambda,
                 reg_alpha=p.reg_alpha,
@@ -260,8 +303,7 @@ class HistGrower:
                     (
                         -gain,
                         counter,
-                        nid,
-                        depth,
+                        node.nid,
                         {
                             "feature": int(s["feature"][0]),
                             "bin": int(s["bin"][0]),
@@ -274,38 +316,52 @@ class HistGrower:
                 )
             counter += 1
 
-        evaluate(0, 0)
+        evaluate(root_node)
         n_leaves = 1
+        split_ids = set()
         while heap and n_leaves < max_leaves:
-            neg_gain, _, nid, depth, s = heapq.heappop(heap)
-            if p.max_depth and depth >= p.max_depth:
+            _neg, _c, nid, s = heapq.heappop(heap)
+            node = nodes[nid]
+            if p.max_depth and node.depth >= p.max_depth:
                 continue
+            counts = self.backend.partition_level(
+                self.qm,
+                self._bufs[node.parity],
+                self._bufs[1 - node.parity],
+                [node.seg],
+                [s["feature"]],
+                [s["bin"]],
+                [s["default_left"]],
+            )
             lid, rid = self._apply_split(
-                tree, nid, s["feature"], s["bin"], s["default_left"], s["gain"],
-                s["left_g"], s["left_h"], node_sum,
+                tree, node, s["feature"], s["bin"], s["default_left"], s["gain"], s["left_g"], s["left_h"]
             )
-            left_rows, right_rows = self.backend.partition_rows(
-                self.qm, node_rows[nid], s["feature"], s["bin"], s["default_left"]
+            mid = node.start + counts[0]
+            lnode = _Node(lid, 1 - node.parity, node.start, mid, s["left_g"], s["left_h"], node.depth + 1)
+            rnode = _Node(
+                rid, 1 - node.parity, mid, node.end, node.g - s["left_g"], node.h - s["left_h"], node.depth + 1
             )
-            node_rows[lid] = left_rows
-            node_rows[rid] = right_rows
-            del node_rows[nid]
+            nodes[lid] = lnode
+            nodes[rid] = rnode
+            split_ids.add(nid)
+            del nodes[nid]
             n_leaves += 1
-            evaluate(lid, depth + 1)
-            evaluate(rid, depth + 1)
+            # build smaller child first so the larger derives by subtraction
+            first, second = (lnode, rnode) if lnode.h <= rnode.h else (rnode, lnode)
+            evaluate(first, second)
+            evaluate(second, first)
             node_hist.pop(nid, None)
         node_hist.clear()
-        return node_rows
+        return [(n.parity, n.start, n.end, float(tree.value[n.nid])) for n in nodes.values()]
 
-    # -- shared ------------------------------------------------------------
-    def _apply_split(self, tree, nid, feature, bin_idx, default_left, gain, left_g, left_h, node_sum):
+    # -- shared ---------------------------------------------------------------
+    def _apply_split(self, tree, node, feature, bin_idx, default_left, gain, left_g, left_h):
         p = self.p
-        G, H = node_sum[nid]
-        GR, HR = G - left_g, H - left_h
+        GR, HR = node.g - left_g, node.h - left_h
         cut_base = int(self.qm.cut_ptr[feature])
         threshold = float(self.qm.cuts[cut_base + bin_idx])
-        lid, rid = tree.apply_split(
-            nid,
+        return tree.apply_split(
+            node.nid,
             feature,
             threshold,
             bin_idx,
@@ -316,6 +372,3 @@ class HistGrower:
             left_hess=left_h,
             right_hess=HR,
         )
-        node_sum[lid] = (left_g, left_h)
-        node_sum[rid] = (GR, HR)
-        return lid, rid

@@ -181,13 +181,8 @@ def train(
         for _parallel in range(num_parallel_tree):
             for cls in range(n_outputs):
                 gh_cls = gh if n_outputs == 1 else gh[:, cls, :].contiguous()
-                tree, leaf_segments = grower.grow(gh_cls)
-                backend.update_margins(
-                    margin[:, cls],
-                    0,
-                    list(leaf_segments.values()),
-                    [float(tree.value[nid]) for nid in leaf_segments],
-                )
+                tree, leaf_jobs = grower.grow(gh_cls)
+                backend.update_margins(margin[:, cls], grower.bufs, leaf_jobs)
                 round_trees.append(tree)
                 round_info.append(cls)
                 for es in eval_sets:

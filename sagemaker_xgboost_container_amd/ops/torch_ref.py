@@ -3,33 +3,48 @@
 This is (a) the CPU execution path for small jobs/tests and (b) the numerics
 reference the CDNA4 HIP kernels are validated against (fp64 histogram
 accumulation). The function contracts here define the backend interface the
-HIP module implements.
+HIP module (ops/hip.py + csrc/) implements.
+
+Interface (segment-based, one device sync per tree level):
+  compute_scale(gh)                     -> backend-specific scale (None here)
+  build_histograms(qm, gh, rowbuf, jobs, scale) -> accumulator [J, slots, 2]
+  hist_to_float(acc, scale)             -> float32 view of the accumulator
+  find_splits(hist, parent_sum, qm, **) -> best split per node
+  partition_level(qm, src, dst, segs, feats, bins, dls) -> left counts (host)
+  update_margins(margin_col, bufs, leaf_jobs)
+  predict_tree(tree, X)                 -> (n,) margin contribution
 """
 import torch
 
 NAME = "torch_ref"
 
 
-def build_histogram(qm, gh, row_idx, out=None):
-    """Accumulate (grad, hess) histograms for one node's rows.
+def compute_scale(gh, comm=None):
+    """torch_ref accumulates in fp64 — no fixed-point scale needed."""
+    return None
 
-    qm: QuantizedMatrix; gh: (n, 2) float32; row_idx: (m,) int32/int64
-    Returns (f * stride, 2) float32 histogram (accumulated in fp64).
+
+def build_histograms(qm, gh, rowbuf, jobs, scale=None):
+    """Accumulate (grad, hess) histograms for a batch of row segments.
+
+    jobs: list of (start, end) into rowbuf. Returns (J, f*stride, 2) fp64.
     """
     f = qm.num_col
     stride = qm.stride
-    rows = row_idx.long()
-    bins = qm.bins[rows].long()  # (m, f)
-    offsets = torch.arange(f, device=bins.device, dtype=torch.long) * stride
-    slots = (bins + offsets).reshape(-1)  # (m*f,)
-    weights = gh[rows].to(torch.float64)  # (m, 2)
-    weights = weights.repeat_interleave(f, dim=0)  # (m*f, 2)
-    hist = torch.zeros((f * stride, 2), dtype=torch.float64, device=bins.device)
-    hist.index_add_(0, slots, weights)
-    if out is not None:
-        out.copy_(hist.to(out.dtype))
-        return out
-    return hist.to(torch.float32)
+    device = qm.bins.device
+    offsets = torch.arange(f, device=device, dtype=torch.long) * stride
+    acc = torch.zeros((len(jobs), f * stride, 2), dtype=torch.float64, device=device)
+    for i, (start, end) in enumerate(jobs):
+        rows = rowbuf[start:end].long()
+        bins = qm.bins[rows].long()  # (m, f)
+        slots = (bins + offsets).reshape(-1)
+        weights = gh[rows].to(torch.float64).repeat_interleave(f, dim=0)
+        acc[i].index_add_(0, slots, weights)
+    return acc
+
+
+def hist_to_float(acc, scale=None):
+    return acc.to(torch.float32)
 
 
 def find_splits(
@@ -57,7 +72,8 @@ def find_splits(
 
     nbins = qm.nbins.to(device)  # (f,)
     bin_ar = torch.arange(stride, device=device)
-    valid_bin = bin_ar.unsqueeze(0) < (nbins.unsqueeze(1) - 1)  # (f, stride): split after bin j needs j < nbins-1
+    # split after bin j is valid iff j <= nbins_f - 2 (right side non-empty)
+    valid_bin = bin_ar.unsqueeze(0) < (nbins.unsqueeze(1) - 1)  # (f, stride)
 
     if qm.has_missing:
         missing = h[:, :, stride - 1, :]  # (k, f, 2)
@@ -122,35 +138,41 @@ def find_splits(
     }
 
 
-def partition_rows(qm, row_idx, feature, split_bin, default_left):
-    """Split one node's rows into (left_rows, right_rows).
-
-    row goes left iff bin <= split_bin (missing: default_left).
-    """
-    rows = row_idx.long()
-    bins = qm.bins[rows, int(feature)].long()
+def _go_left_mask(qm, bins, split_bin, default_left):
     if qm.has_missing:
         is_missing = bins == (qm.stride - 1)
-        go_left = torch.where(
+        return torch.where(
             is_missing,
             torch.full_like(is_missing, bool(default_left)),
             bins <= int(split_bin),
         )
-    else:
-        go_left = bins <= int(split_bin)
-    return row_idx[go_left], row_idx[~go_left]
+    return bins <= int(split_bin)
 
 
-def update_margins(margin, out_col, row_idx_segments, leaf_values):
-    """margin[rows, out_col] += leaf_value for each (rows, value) segment."""
-    for rows, value in zip(row_idx_segments, leaf_values):
-        if margin.dim() == 1:
-            margin[rows.long()] += value
-        else:
-            margin[rows.long(), out_col] += value
+def partition_level(qm, src, dst, segs, feats, split_bins, default_lefts):
+    """Partition each (start, end) segment of src into dst (left block then
+    right block in the same index range). Returns list of left counts."""
+    counts = []
+    for (start, end), feature, sbin, dl in zip(segs, feats, split_bins, default_lefts):
+        rows = src[start:end]
+        bins = qm.bins[rows.long(), int(feature)].long()
+        go_left = _go_left_mask(qm, bins, sbin, dl)
+        left = rows[go_left]
+        right = rows[~go_left]
+        dst[start : start + left.numel()] = left
+        dst[start + left.numel() : end] = right
+        counts.append(int(left.numel()))
+    return counts
 
 
-def predict_tree(tree, X, missing_nan=True):
+def update_margins(margin_col, bufs, leaf_jobs):
+    """margin_col[rows] += value for each (parity, start, end, value) job."""
+    for parity, start, end, value in leaf_jobs:
+        rows = bufs[parity][start:end].long()
+        margin_col[rows] += value
+
+
+def predict_tree(tree, X):
     """Margin contribution of one tree for dense X (n, f) float32 (NaN missing).
 
     tree: models.tree.Tree (host arrays). Vectorized level-by-level traversal.
