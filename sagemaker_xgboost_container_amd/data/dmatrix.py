@@ -285,3 +285,42 @@ class DMatrix:
     def __repr__(self):
         kind = "sparse" if self.is_sparse else "dense"
         return f"DMatrix({self.num_row()}x{self.num_col()}, {kind})"
+
+
+class DeviceDMatrix:
+    """DMatrix backed by tensors that already live on the training device.
+
+    Used for GPU-resident synthetic data (bench.py) and any caller that
+    builds features directly in HBM — avoids the host round-trip of the
+    numpy-backed DMatrix. Exposes the subset of the DMatrix surface the
+    trainer consumes.
+    """
+
+    def __init__(self, data, label=None, weight=None, feature_names=None, base_margin=None):
+        self._data = data
+        self._label = label
+        self._weight = weight
+        self._base_margin = base_margin
+        self.feature_names = feature_names
+
+    def num_row(self):
+        return self._data.shape[0]
+
+    def num_col(self):
+        return self._data.shape[1]
+
+    def to_dense(self):
+        return self._data
+
+    def get_label(self):
+        import torch
+
+        return self._label if self._label is not None else torch.zeros(0)
+
+    def get_weight(self):
+        import torch
+
+        return self._weight if self._weight is not None else torch.zeros(0)
+
+    def get_base_margin(self):
+        return self._base_margin

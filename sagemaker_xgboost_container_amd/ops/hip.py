@@ -1,0 +1,189 @@
+"""CDNA4 HIP ops backend (MI355X compute path).
+
+Wraps the in-tree extension built from csrc/smxgb_kernels.hip. Implements
+the same segment-based interface as ops/torch_ref.py. Host code here only
+PACKS work descriptors (job tables, block->job maps) — all per-row compute
+runs in the HIP kernels.
+
+This module raises ImportError loudly when the extension is missing: on a
+GPU box the HIP path must run — there is no silent eager fallback
+(set SMXGB_FORCE_TORCH_OPS=1 explicitly for ablation).
+"""
+import numpy as np
+import torch
+
+from . import torch_ref
+
+NAME = "hip"
+
+try:
+    from . import _smxgb_hip as _K  # built in-tree by setup.py build_ext --inplace
+except ImportError as _e:  # pragma: no cover - exercised only on GPU boxes
+    raise ImportError(
+        "smxgb HIP extension is not built. Run `python setup.py build_ext --inplace` "
+        f"(hipcc, PYTORCH_ROCM_ARCH=gfx950). Underlying error: {_e}"
+    )
+
+# fixed-point scale: values mapped to v * 2^33 / max_abs; sums of up to ~5e8
+# rows stay inside int64 with 2^29 headroom.
+_FIXED_BITS = 33.0
+
+# LDS histogram slab per feature group (56 KiB -> 2 blocks/CU co-residency)
+_LDS_BYTES = 56 * 1024
+_ROWS_PER_BLOCK = 8192
+_MAX_BLOCKS_PER_JOB = 512
+
+
+def compute_scale(gh, comm=None):
+    m = gh.abs().amax(dim=0)  # (2,)
+    if comm is not None:
+        comm.allreduce_max_(m)
+    gmax, hmax = (float(v) for v in m.cpu())
+    scale_g = 2.0**_FIXED_BITS / max(gmax, 1e-30)
+    scale_h = 2.0**_FIXED_BITS / max(hmax, 1e-30)
+    return (scale_g, scale_h)
+
+
+def _feature_groups(nfeat, stride):
+    per_group = max(1, min(nfeat, (_LDS_BYTES // 16) // max(stride, 1)))
+    groups = []
+    f = 0
+    while f < nfeat:
+        groups.append((f, min(f + per_group, nfeat)))
+        f += per_group
+    return groups
+
+
+def _pack_jobs(fields_list):
+    """fields_list: list of tuples of int32 words -> (jobs_dev, width)."""
+    arr = np.asarray(fields_list, dtype=np.int32)
+    return torch.from_numpy(arr).cuda(non_blocking=True)
+
+
+def _block_map(blocks_per_job):
+    total = int(sum(blocks_per_job))
+    bj = np.empty(total, dtype=np.int32)
+    ofs = 0
+    for j, nb in enumerate(blocks_per_job):
+        bj[ofs : ofs + nb] = j
+        ofs += nb
+    return torch.from_numpy(bj).cuda(non_blocking=True)
+
+
+def build_histograms(qm, gh, rowbuf, jobs, scale):
+    f = qm.num_col
+    stride = qm.stride
+    groups = _feature_groups(f, stride)
+    acc = torch.zeros((len(jobs), f * stride, 2), dtype=torch.int64, device=qm.bins.device)
+
+    job_rows = []
+    first_block = 0
+    packed = []
+    blocks_per = []
+    for hist_idx, (start, end) in enumerate(jobs):
+        rows = end - start
+        nb = int(max(1, min((rows + _ROWS_PER_BLOCK - 1) // _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB)))
+        for fg_start, fg_end in groups:
+            packed.append((start, end, hist_idx, fg_start, fg_end, first_block, nb))
+            blocks_per.append(nb)
+            first_block += nb
+        job_rows.append(rows)
+
+    jobs_dev = _pack_jobs(packed)
+    block_job = _block_map(blocks_per)
+    lds_words = max((fe - fs) for fs, fe in groups) * stride * 2
+    _K.hist_build(
+        qm.bins, gh.contiguous(), rowbuf, jobs_dev, block_job, acc,
+        f, stride, scale[0], scale[1], lds_words,
+    )
+    return acc
+
+
+def hist_to_float(acc, scale):
+    out = torch.empty(acc.shape, dtype=torch.float32, device=acc.device)
+    _K.hist_convert(acc, out, 1.0 / scale[0], 1.0 / scale[1])
+    return out
+
+
+# split scan operates on (nodes x total_bins) tensors — small; the torch
+# implementation runs as a handful of fused kernels on-device.
+find_splits = torch_ref.find_splits
+
+
+def partition_level(qm, src, dst, segs, feats, split_bins, default_lefts):
+    J = len(segs)
+    packed = []
+    blocks_per = []
+    first_block = 0
+    for (start, end), feature, sbin, dl in zip(segs, feats, split_bins, default_lefts):
+        rows = end - start
+        nb = int(max(1, min((rows + _ROWS_PER_BLOCK - 1) // _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB)))
+        packed.append((start, end, int(feature), int(sbin), int(bool(dl)), first_block, nb))
+        blocks_per.append(nb)
+        first_block += nb
+    jobs_dev = _pack_jobs(packed)
+    block_job = _block_map(blocks_per)
+    counters = torch.zeros((J, 2), dtype=torch.int32, device=src.device)
+    missing_bin = qm.stride - 1 if qm.has_missing else -1
+    _K.partition(qm.bins, src, dst, jobs_dev, block_job, counters, qm.num_col, missing_bin)
+    return counters[:, 0].cpu().tolist()  # the level's single device sync
+
+
+def update_margins(margin_col, bufs, leaf_jobs):
+    if not leaf_jobs:
+        return
+    packed = []
+    blocks_per = []
+    first_block = 0
+    for parity, start, end, value in leaf_jobs:
+        rows = end - start
+        nb = int(max(1, min((rows + _ROWS_PER_BLOCK - 1) // _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB)))
+        vbits = int(np.float32(value).view(np.int32))
+        packed.append((start, end, int(parity), vbits, first_block, nb))
+        blocks_per.append(nb)
+        first_block += nb
+    jobs_dev = _pack_jobs(packed)
+    block_job = _block_map(blocks_per)
+    _K.leaf_update(bufs[0], bufs[1], margin_col, jobs_dev, block_job, margin_col.stride(0))
+
+
+class _FlatForest:
+    """Trees flattened into device arrays with global node ids."""
+
+    def __init__(self, trees, tree_info, device):
+        import numpy as np
+
+        offsets = np.cumsum([0] + [t.num_nodes for t in trees]).astype(np.int32)
+        left = np.concatenate([t.left + (t.left >= 0) * offsets[i] for i, t in enumerate(trees)])
+        right = np.concatenate([t.right + (t.right >= 0) * offsets[i] for i, t in enumerate(trees)])
+        self.left = torch.from_numpy(left.astype(np.int32)).to(device)
+        self.right = torch.from_numpy(right.astype(np.int32)).to(device)
+        self.feat = torch.from_numpy(np.concatenate([t.feature for t in trees]).astype(np.int32)).to(device)
+        self.thresh = torch.from_numpy(np.concatenate([t.threshold for t in trees]).astype(np.float32)).to(device)
+        self.defl = torch.from_numpy(
+            np.concatenate([t.default_left for t in trees]).astype(np.uint8)
+        ).to(device)
+        self.value = torch.from_numpy(np.concatenate([t.value for t in trees]).astype(np.float32)).to(device)
+        self.tree_root = torch.from_numpy(offsets[:-1]).to(device)
+        self.tree_cls = torch.from_numpy(np.asarray(tree_info, dtype=np.int32)).to(device)
+        self.n_trees = len(trees)
+
+
+def predict_forest(trees, tree_info, X, k, t_begin=0, t_end=None, out=None):
+    """Summed margin contributions of trees[t_begin:t_end] -> (n, k)."""
+    forest = _FlatForest(trees, tree_info, X.device)
+    n = X.shape[0]
+    if out is None:
+        out = torch.zeros((n, k), dtype=torch.float32, device=X.device)
+    if t_end is None:
+        t_end = forest.n_trees
+    _K.predict_forest(
+        X.contiguous(), forest.left, forest.right, forest.feat, forest.thresh, forest.defl,
+        forest.value, forest.tree_root, forest.tree_cls, t_begin, t_end, out, k,
+    )
+    return out
+
+
+def predict_tree(tree, X):
+    out = predict_forest([tree], [0], X, 1)
+    return out[:, 0]

@@ -1,0 +1,210 @@
+"""HIP kernel numerics vs the plain-PyTorch fp32/fp64 reference backend.
+
+All tests require an MI355X (run via gpurun: pytest tests -m gpu).
+"""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from sagemaker_xgboost_container_amd.ops import hip, torch_ref
+    from sagemaker_xgboost_container_amd.ops.quantize import quantize
+else:  # allow collection on CPU boxes
+    hip = torch_ref = quantize = None
+
+
+def _random_problem(n=200_000, f=28, max_bin=256, missing=False, seed=0):
+    rng = np.random.default_rng(seed)
+    X = rng.normal(size=(n, f)).astype(np.float32)
+    if missing:
+        X[rng.random(size=X.shape) < 0.05] = np.nan
+    Xg = torch.tensor(X, device="cuda")
+    qm = quantize(Xg, max_bin=max_bin)
+    gh = torch.stack(
+        [
+            torch.tensor(rng.normal(size=n).astype(np.float32), device="cuda"),
+            torch.tensor(rng.random(size=n).astype(np.float32) + 0.01, device="cuda"),
+        ],
+        dim=1,
+    )
+    return qm, gh
+
+
+@pytest.fixture(scope="module")
+def problem():
+    return _random_problem()
+
+
+class TestHistogram:
+    def test_matches_reference(self, problem):
+        qm, gh = problem
+        n = qm.num_row
+        rowbuf = torch.arange(n, dtype=torch.int32, device="cuda")
+        jobs = [(0, n // 2), (n // 2, n)]
+        scale = hip.compute_scale(gh)
+        acc = hip.build_histograms(qm, gh, rowbuf, jobs, scale)
+        hist_hip = hip.hist_to_float(acc, scale)
+        hist_ref = torch_ref.hist_to_float(torch_ref.build_histograms(qm, gh, rowbuf, jobs, None))
+        torch.cuda.synchronize()
+        ref = hist_ref.cpu().numpy()
+        got = hist_hip.cpu().numpy()
+        denom = np.abs(ref).max()
+        np.testing.assert_allclose(got, ref, atol=denom * 1e-5)
+
+    def test_deterministic(self, problem):
+        qm, gh = problem
+        n = qm.num_row
+        rowbuf = torch.arange(n, dtype=torch.int32, device="cuda")
+        scale = hip.compute_scale(gh)
+        a1 = hip.build_histograms(qm, gh, rowbuf, [(0, n)], scale)
+        a2 = hip.build_histograms(qm, gh, rowbuf, [(0, n)], scale)
+        torch.cuda.synchronize()
+        assert torch.equal(a1, a2), "fixed-point histogram must be bit-deterministic"
+
+    def test_missing_bins(self):
+        qm, gh = _random_problem(n=50_000, f=10, missing=True, seed=1)
+        assert qm.has_missing
+        n = qm.num_row
+        rowbuf = torch.arange(n, dtype=torch.int32, device="cuda")
+        scale = hip.compute_scale(gh)
+        hist_hip = hip.hist_to_float(hip.build_histograms(qm, gh, rowbuf, [(0, n)], scale), scale)
+        hist_ref = torch_ref.hist_to_float(torch_ref.build_histograms(qm, gh, rowbuf, [(0, n)], None))
+        ref = hist_ref.cpu().numpy()
+        np.testing.assert_allclose(hist_hip.cpu().numpy(), ref, atol=np.abs(ref).max() * 1e-5)
+
+    def test_int16_bins(self):
+        # >256 slots forces the int16 bin path (max_bin 256 + missing)
+        qm, gh = _random_problem(n=50_000, f=8, max_bin=256, missing=True, seed=2)
+        if qm.bins.dtype == torch.uint8:
+            pytest.skip("stride fits uint8")
+        n = qm.num_row
+        rowbuf = torch.arange(n, dtype=torch.int32, device="cuda")
+        scale = hip.compute_scale(gh)
+        hist_hip = hip.hist_to_float(hip.build_histograms(qm, gh, rowbuf, [(0, n)], scale), scale)
+        hist_ref = torch_ref.hist_to_float(torch_ref.build_histograms(qm, gh, rowbuf, [(0, n)], None))
+        ref = hist_ref.cpu().numpy()
+        np.testing.assert_allclose(hist_hip.cpu().numpy(), ref, atol=np.abs(ref).max() * 1e-5)
+
+
+class TestPartition:
+    def test_matches_reference_sets(self, problem):
+        qm, _gh = problem
+        n = qm.num_row
+        src = torch.arange(n, dtype=torch.int32, device="cuda")
+        dst = torch.empty_like(src)
+        segs = [(0, n // 3), (n // 3, n)]
+        feats = [0, 5]
+        sbins = [100, 37]
+        dls = [False, True]
+        counts = hip.partition_level(qm, src, dst, segs, feats, sbins, dls)
+
+        src_ref = src.clone()
+        dst_ref = torch.empty_like(src_ref)
+        counts_ref = torch_ref.partition_level(qm, src_ref, dst_ref, segs, feats, sbins, dls)
+        assert counts == counts_ref
+        for (start, end), lc in zip(segs, counts):
+            left_hip = set(dst[start : start + lc].cpu().tolist())
+            left_ref = set(dst_ref[start : start + lc].cpu().tolist())
+            assert left_hip == left_ref
+            right_hip = set(dst[start + lc : end].cpu().tolist())
+            right_ref = set(dst_ref[start + lc : end].cpu().tolist())
+            assert right_hip == right_ref
+
+
+class TestLeafUpdate:
+    def test_scatter(self):
+        n = 10_000
+        margin = torch.zeros((n, 2), dtype=torch.float32, device="cuda")
+        buf0 = torch.randperm(n, device="cuda").to(torch.int32)
+        buf1 = torch.randperm(n, device="cuda").to(torch.int32)
+        jobs = [(0, 0, n // 2, 0.5), (1, n // 2, n, -0.25)]
+        hip.update_margins(margin[:, 1], (buf0, buf1), jobs)
+        ref = torch.zeros((n, 2), dtype=torch.float32, device="cuda")
+        torch_ref.update_margins(ref[:, 1], (buf0, buf1), jobs)
+        torch.cuda.synchronize()
+        assert torch.allclose(margin, ref)
+        assert margin[:, 0].abs().sum() == 0
+
+
+class TestPredict:
+    def test_forest_matches_reference(self):
+        from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+        from sagemaker_xgboost_container_amd.models import trainer
+
+        rng = np.random.default_rng(3)
+        X = rng.normal(size=(5000, 12)).astype(np.float32)
+        X[rng.random(size=X.shape) < 0.03] = np.nan
+        y = (np.nan_to_num(X[:, 0]) > 0).astype(np.float32)
+        bst = trainer.train(
+            {"objective": "binary:logistic", "max_depth": 5, "device": "cpu"},
+            DMatrix(X, label=y),
+            num_boost_round=5,
+            verbose_eval=False,
+        )
+        Xg = torch.tensor(X, device="cuda")
+        out_hip = hip.predict_forest(bst.trees, bst.tree_info, Xg, 1)
+        ref = torch.zeros((X.shape[0], 1), device="cuda")
+        for t in bst.trees:
+            ref[:, 0] += torch_ref.predict_tree(t, Xg)
+        torch.cuda.synchronize()
+        np.testing.assert_allclose(out_hip.cpu().numpy(), ref.cpu().numpy(), rtol=1e-5, atol=1e-6)
+
+
+class TestEndToEnd:
+    def test_gpu_training_learns(self):
+        from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+        from sagemaker_xgboost_container_amd.models import trainer
+
+        rng = np.random.default_rng(4)
+        X = rng.normal(size=(100_000, 28)).astype(np.float32)
+        logit = X[:, 0] * 2 - X[:, 1] + 0.5 * X[:, 2] * X[:, 3]
+        y = (logit + rng.normal(scale=0.5, size=len(X)) > 0).astype(np.float32)
+        res = {}
+        bst = trainer.train(
+            {"objective": "binary:logistic", "max_depth": 6, "eta": 0.3, "tree_method": "gpu_hist"},
+            DMatrix(X, label=y),
+            num_boost_round=10,
+            evals=[(DMatrix(X, label=y), "train")],
+            evals_result=res,
+            verbose_eval=False,
+        )
+        assert res["train"]["logloss"][-1] < 0.35
+        pred = bst.predict(X[:1000])
+        assert ((pred > 0.5) == y[:1000]).mean() > 0.85
+
+    def test_gpu_matches_cpu_model(self):
+        """Same small dataset: GPU-grown model close to CPU-grown model."""
+        from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+        from sagemaker_xgboost_container_amd.models import trainer
+
+        rng = np.random.default_rng(5)
+        X = rng.normal(size=(20_000, 10)).astype(np.float32)
+        y = (X[:, 0] + 0.5 * X[:, 1] > 0).astype(np.float32)
+        params = {"objective": "binary:logistic", "max_depth": 4, "eta": 0.5}
+        cpu = trainer.train({**params, "device": "cpu"}, DMatrix(X, label=y), 5, verbose_eval=False)
+        gpu = trainer.train({**params, "device": "cuda"}, DMatrix(X, label=y), 5, verbose_eval=False)
+        p_cpu = cpu.predict(X[:2000])
+        p_gpu = gpu.predict(X[:2000])
+        # identical cuts + deterministic hist => same trees up to fp noise
+        assert np.mean(np.abs(p_cpu - p_gpu)) < 5e-3
+
+    def test_multiclass_gpu(self):
+        from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+        from sagemaker_xgboost_container_amd.models import trainer
+
+        rng = np.random.default_rng(6)
+        X = rng.normal(size=(50_000, 54)).astype(np.float32)
+        y = ((X[:, 0] > 0.5).astype(int) + (X[:, 1] > 0).astype(int) * 3).astype(np.float32)
+        y = np.clip(y, 0, 6)
+        res = {}
+        trainer.train(
+            {"objective": "multi:softprob", "num_class": 7, "max_depth": 6, "tree_method": "gpu_hist"},
+            DMatrix(X, label=y),
+            num_boost_round=3,
+            evals=[(DMatrix(X, label=y), "train")],
+            evals_result=res,
+            verbose_eval=False,
+        )
+        assert res["train"]["mlogloss"][-1] < res["train"]["mlogloss"][0]
