@@ -58,7 +58,11 @@ class _EvalSet:
         self.X = torch.as_tensor(dmatrix.to_dense(), dtype=torch.float32, device=device)
         self.y = torch.as_tensor(dmatrix.get_label(), dtype=torch.float32, device=device)
         w = dmatrix.get_weight()
-        self.w = torch.as_tensor(w, dtype=torch.float32, device=device) if w.size else None
+        self.w = (
+            torch.as_tensor(w, dtype=torch.float32, device=device)
+            if w is not None and len(w)
+            else None
+        )
         self.dmatrix = dmatrix
         user_margin = dmatrix.get_base_margin()
         if user_margin is not None:
@@ -120,8 +124,12 @@ def train(
     # -- resident training state ------------------------------------------
     X = torch.as_tensor(dtrain.to_dense(), dtype=torch.float32, device=device)
     y = torch.as_tensor(dtrain.get_label(), dtype=torch.float32, device=device)
-    w_np = dtrain.get_weight()
-    weight = torch.as_tensor(w_np, dtype=torch.float32, device=device) if w_np.size else None
+    w_raw = dtrain.get_weight()
+    weight = (
+        torch.as_tensor(w_raw, dtype=torch.float32, device=device)
+        if w_raw is not None and len(w_raw)
+        else None
+    )
     objective.validate_labels(y)
 
     max_bin = int(params.get("max_bin", 256))
