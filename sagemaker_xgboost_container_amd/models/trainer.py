@@ -133,7 +133,7 @@ def train(
     objective.validate_labels(y)
 
     max_bin = int(params.get("max_bin", 256))
-    qm = quantize(X, max_bin=max_bin, sample_weight=weight)
+    qm = quantize(X, max_bin=max_bin, sample_weight=weight, comm=comm)
     backend = backend_for(device)
 
     base_margin_value = objective.base_margin(booster.base_score)

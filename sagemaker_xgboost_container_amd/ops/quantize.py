@@ -90,6 +90,86 @@ def _feature_cuts(values, max_bin, weights=None):
     return torch.unique(cand)
 
 
+def make_cuts_distributed(X, max_bin=256, sample_weight=None, comm=None, n_candidates=None):
+    """Globally consistent cuts across ranks (replaces the distributed
+    quantile-sketch allreduce inside xgboost).
+
+    Each rank summarizes every feature as `n_candidates` values at uniform
+    positions of its local (weighted) CDF, each carrying weight
+    local_mass / n_candidates. Summaries are allgathered (ONE fused
+    collective) and the merged weighted multiset is re-quantiled into at
+    most max_bin - 1 cuts — identical on every rank by construction.
+    """
+    n, f = X.shape
+    ncand = n_candidates or min(4 * max_bin, 4096)
+    device = X.device
+
+    cand = torch.full((f, ncand), float("nan"), device=device)
+    mass = torch.zeros(f, device=device)
+    for j in range(f):
+        col = X[:, j]
+        finite_mask = ~torch.isnan(col)
+        finite = col[finite_mask]
+        if finite.numel() == 0:
+            continue
+        sorted_vals, order = torch.sort(finite)
+        if sample_weight is not None:
+            w = sample_weight[finite_mask][order]
+            cw = torch.cumsum(w, 0)
+            total = cw[-1]
+            targets = (torch.arange(ncand, device=device, dtype=torch.float32) + 0.5) / ncand * total
+            pos = torch.searchsorted(cw, targets).clamp_(0, finite.numel() - 1)
+            cand[j] = sorted_vals[pos]
+            mass[j] = total
+        else:
+            pos = ((torch.arange(ncand, device=device, dtype=torch.float64) + 0.5) / ncand * finite.numel()).long()
+            cand[j] = sorted_vals[pos.clamp_(0, finite.numel() - 1)]
+            mass[j] = float(finite.numel())
+
+    if comm is not None and comm.world_size > 1:
+        world = comm.world_size
+        all_cand = torch.zeros((world, f, ncand), device=device)
+        all_mass = torch.zeros((world, f), device=device)
+        all_cand[comm.rank] = cand
+        all_mass[comm.rank] = mass
+        comm.allreduce_(all_cand)
+        comm.allreduce_(all_mass)
+    else:
+        all_cand = cand.unsqueeze(0)
+        all_mass = mass.unsqueeze(0)
+
+    cut_list = []
+    nbins = torch.empty(f, dtype=torch.int64)
+    for j in range(f):
+        values = all_cand[:, j, :].reshape(-1)
+        weights = (all_mass[:, j] / ncand).reshape(-1, 1).expand(-1, ncand).reshape(-1)
+        # drop NaN slots and the zero-weight padding contributed by the
+        # allreduce-based gather (ranks write only their own slice)
+        keep = (~torch.isnan(values)) & (weights > 0)
+        values, weights = values[keep], weights[keep]
+        if values.numel() == 0:
+            cut_list.append(X.new_zeros((0,)))
+            nbins[j] = 1
+            continue
+        distinct = torch.unique(values)
+        if distinct.numel() <= max_bin:
+            cuts_j = (distinct[:-1] + distinct[1:]) * 0.5 if distinct.numel() > 1 else X.new_zeros((0,))
+        else:
+            sorted_vals, order = torch.sort(values)
+            cw = torch.cumsum(weights[order], 0)
+            total = cw[-1]
+            targets = torch.linspace(0, 1, max_bin + 1, device=device)[1:-1] * total
+            pos = torch.searchsorted(cw, targets).clamp_(0, values.numel() - 1)
+            cuts_j = torch.unique(sorted_vals[pos])
+        cut_list.append(cuts_j)
+        nbins[j] = cuts_j.numel() + 1
+
+    cut_ptr = torch.zeros(f + 1, dtype=torch.int64)
+    cut_ptr[1:] = torch.cumsum(torch.tensor([c.numel() for c in cut_list]), 0)
+    cuts_flat = torch.cat(cut_list) if cut_list else X.new_zeros((0,))
+    return cuts_flat, cut_ptr, nbins.to(device)
+
+
 def make_cuts(X, max_bin=256, sample_weight=None):
     """Per-feature cuts from a dense float32 matrix with NaN missing.
 
@@ -113,17 +193,29 @@ def make_cuts(X, max_bin=256, sample_weight=None):
     return cuts_flat, cut_ptr, nbins.to(X.device)
 
 
-def quantize(X, max_bin=256, sample_weight=None, cuts=None, cut_ptr=None, nbins=None):
+def quantize(X, max_bin=256, sample_weight=None, cuts=None, cut_ptr=None, nbins=None, comm=None):
     """Quantize dense X (n, f) float32 with NaN missing into a QuantizedMatrix.
 
     Pass precomputed (cuts, cut_ptr, nbins) to bin an eval/serving matrix
-    with training cuts.
+    with training cuts. With a communicator, cuts are computed from the
+    GLOBAL distribution (distributed sketch merge) so every rank bins
+    identically.
     """
     n, f = X.shape
     if cuts is None:
-        cuts, cut_ptr, nbins = make_cuts(X, max_bin=max_bin, sample_weight=sample_weight)
+        if comm is not None and comm.world_size > 1:
+            cuts, cut_ptr, nbins = make_cuts_distributed(
+                X, max_bin=max_bin, sample_weight=sample_weight, comm=comm
+            )
+        else:
+            cuts, cut_ptr, nbins = make_cuts(X, max_bin=max_bin, sample_weight=sample_weight)
 
     has_missing = bool(torch.isnan(X).any().item())
+    if comm is not None and comm.world_size > 1:
+        # stride/missing-slot layout must agree across ranks
+        flag = torch.tensor([1.0 if has_missing else 0.0], device=X.device)
+        comm.allreduce_max_(flag)
+        has_missing = bool(flag.item() > 0)
     max_nbins = int(nbins.max().item()) if f else 1
     stride = max_nbins + (1 if has_missing else 0)
     dtype = torch.uint8 if stride <= 256 else torch.int16
