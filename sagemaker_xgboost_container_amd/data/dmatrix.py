@@ -259,10 +259,18 @@ class DMatrix:
 
     # -- views ------------------------------------------------------------
     def to_dense(self):
-        """Dense float32 view (NaN = missing). Materializes sparse data."""
+        """Dense float32 view (NaN = missing). Materializes sparse data.
+
+        Entries ABSENT from a sparse matrix are missing (NaN), matching
+        xgboost's sparse semantics — explicit stored zeros stay 0.
+        """
         if self._dense is not None:
             return self._dense
-        dense = np.asarray(self._csr.todense(), dtype=np.float32)
+        csr = self._csr
+        n, f = csr.shape
+        dense = np.full((n, f), np.nan, dtype=np.float32)
+        rows = np.repeat(np.arange(n), np.diff(csr.indptr))
+        dense[rows, csr.indices] = csr.data
         return dense
 
     def csr(self):

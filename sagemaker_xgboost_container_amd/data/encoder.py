@@ -1,0 +1,98 @@
+"""Inference payload decoders: request body -> DMatrix.
+
+Parity: reference encoder.py:31-142 (csv/libsvm/recordio-protobuf decoders,
+json_to_jsonlines, MIME-dispatching decode).
+"""
+import csv as csv_module
+import io
+import json
+import logging
+
+import numpy as np
+
+from ..constants import xgb_content_types
+from ..toolkit import exceptions as exc
+from .dmatrix import DMatrix
+from .recordio_protobuf import read_recordio_protobuf
+
+_MIME_CSV = "text/csv"
+
+
+def _clean_csv_string(csv_string, delimiter):
+    return ["nan" if x == "" else x for x in csv_string.split(delimiter)]
+
+
+def csv_to_dmatrix(input, dtype=None):
+    """CSV payload (no label column) -> DMatrix. Empty fields become NaN."""
+    csv_string = input.decode() if isinstance(input, bytes) else input
+    sniff_delimiter = csv_module.Sniffer().sniff(csv_string.split("\n")[0][:512]).delimiter
+    delimiter = "," if sniff_delimiter.isalnum() else sniff_delimiter
+    logging.info("Determined delimiter of CSV input is '%s'", delimiter)
+    rows = [_clean_csv_string(line, delimiter) for line in csv_string.split("\n") if line.strip()]
+    np_payload = np.array(rows).astype(dtype or np.float32)
+    return DMatrix(np_payload)
+
+
+def libsvm_to_dmatrix(string_like):
+    """LIBSVM payload (no labels expected) -> dense DMatrix.
+
+    Standard 1-based indices are shifted to 0-based when no index 0 appears.
+    """
+    if isinstance(string_like, (bytes, bytearray)):
+        string_like = string_like.decode("utf-8")
+
+    rows = []
+    for line in string_like.strip().split("\n"):
+        row = {}
+        for token in line.strip().split():
+            if ":" in token:
+                idx, _, val = token.partition(":")
+                row[int(idx)] = float(val)
+        rows.append(row)
+
+    if not rows or not any(rows):
+        return DMatrix(np.empty((0, 0), dtype=np.float32))
+
+    min_idx = min(idx for row in rows for idx in row)
+    offset = 1 if min_idx >= 1 else 0
+    max_col = max(idx for row in rows for idx in row) - offset + 1
+    data = np.zeros((len(rows), max_col), dtype=np.float32)
+    for i, row in enumerate(rows):
+        for idx, val in row.items():
+            data[i, idx - offset] = val
+    return DMatrix(data)
+
+
+def recordio_protobuf_to_dmatrix(string_like):
+    features, labels = read_recordio_protobuf(bytes(string_like))
+    return DMatrix(features, label=labels)
+
+
+_dmatrix_decoders_map = {
+    _MIME_CSV: csv_to_dmatrix,
+    xgb_content_types.LIBSVM: libsvm_to_dmatrix,
+    xgb_content_types.X_LIBSVM: libsvm_to_dmatrix,
+    xgb_content_types.X_RECORDIO_PROTOBUF: recordio_protobuf_to_dmatrix,
+}
+
+
+def json_to_jsonlines(json_data):
+    """{'key': [entries...]} -> one JSON entry per line (bytes)."""
+    resp_dict = json_data if isinstance(json_data, dict) else json.loads(json_data)
+    if len(resp_dict.keys()) != 1:
+        raise ValueError("JSON response is not compatible for conversion to jsonlines.")
+    bio = io.BytesIO()
+    for value in resp_dict.values():
+        for entry in value:
+            bio.write(bytes(json.dumps(entry) + "\n", "UTF-8"))
+    return bio.getvalue()
+
+
+def decode(obj, content_type):
+    """Decode a request payload into a DMatrix by MIME content type."""
+    media_content_type = content_type.split(";")[0].strip().lower()
+    try:
+        decoder = _dmatrix_decoders_map[media_content_type]
+    except KeyError:
+        raise exc.UserError(f"Unsupported content type: {media_content_type}")
+    return decoder(obj)

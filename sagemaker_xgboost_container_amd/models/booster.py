@@ -321,6 +321,7 @@ class Booster:
             {
                 "learner": {
                     "learner_train_param": {"objective": self.objective_name},
+                    "objective": self._objective_json(),
                     "learner_model_param": {
                         "num_class": str(self.num_class),
                         "base_score": repr(self.base_score),

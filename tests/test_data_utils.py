@@ -92,7 +92,9 @@ class TestDMatrixBuild:
         dm = du.get_dmatrix(libsvm_dir, "libsvm")
         assert dm.num_row() == 2 and dm.num_col() == 3
         dense = dm.to_dense()
-        np.testing.assert_allclose(dense[0], [0.5, 0.0, 3.0])
+        # absent sparse entries are missing (NaN), matching xgboost semantics
+        np.testing.assert_allclose(dense[0, [0, 2]], [0.5, 3.0])
+        assert np.isnan(dense[0, 1])
 
     def test_parquet(self, tmp_path):
         import pandas as pd
