@@ -48,14 +48,16 @@ class GrowParams:
         self.colsample_bytree = float(p.get("colsample_bytree", 1.0))
         self.colsample_bylevel = float(p.get("colsample_bylevel", 1.0))
         self.colsample_bynode = float(p.get("colsample_bynode", 1.0))
+        self.monotone_constraints = p.get("monotone_constraints")
+        self.interaction_constraints = p.get("interaction_constraints")
         if self.max_depth == 0 and self.max_leaves == 0 and self.grow_policy == "depthwise":
             self.max_depth = 6
 
 
 class _Node:
-    __slots__ = ("nid", "parity", "start", "end", "g", "h", "depth")
+    __slots__ = ("nid", "parity", "start", "end", "g", "h", "depth", "lower", "upper", "allowed")
 
-    def __init__(self, nid, parity, start, end, g, h, depth):
+    def __init__(self, nid, parity, start, end, g, h, depth, lower=-float("inf"), upper=float("inf"), allowed=None):
         self.nid = nid
         self.parity = parity
         self.start = start
@@ -63,6 +65,9 @@ class _Node:
         self.g = g
         self.h = h
         self.depth = depth
+        self.lower = lower   # monotone-constraint weight bounds
+        self.upper = upper
+        self.allowed = allowed  # interaction-constraint feature mask (f,) bool or None
 
     @property
     def seg(self):
@@ -78,16 +83,50 @@ class HistGrower:
         self.device = qm.device
         self.generator = generator
         self._bufs = None
+        f = qm.num_col
+        self.monotone = None
+        if self.p.monotone_constraints:
+            mono = list(self.p.monotone_constraints)[:f] + [0] * max(0, f - len(self.p.monotone_constraints))
+            if any(mono):
+                self.monotone = torch.tensor(mono, dtype=torch.int8, device=self.device)
+        self.inter_sets = None
+        if self.p.interaction_constraints:
+            self.inter_sets = [set(group) for group in self.p.interaction_constraints]
 
     # -- weight / gain math (host scalars; double precision) ---------------
-    def _weight(self, g, h):
+    def _weight(self, g, h, lower=-float("inf"), upper=float("inf")):
         a = abs(g) - self.p.reg_alpha
         if a < 0:
             a = 0.0
         w = -(a if g > 0 else -a) / (h + self.p.reg_lambda)
         if self.p.max_delta_step > 0:
             w = max(-self.p.max_delta_step, min(self.p.max_delta_step, w))
-        return w
+        return max(lower, min(upper, w))
+
+    def _interaction_allowed(self, parent_allowed, feature):
+        """Child feature mask after splitting on `feature` (xgboost rule:
+        child candidates = parent candidates ∩ (∪ sets containing feature,
+        plus the feature itself))."""
+        if self.inter_sets is None:
+            return None
+        f = self.qm.num_col
+        allowed = torch.zeros(f, dtype=torch.bool, device=self.device)
+        allowed[feature] = True
+        for group in self.inter_sets:
+            if feature in group:
+                for idx in group:
+                    if idx < f:
+                        allowed[idx] = True
+        if parent_allowed is not None:
+            allowed &= parent_allowed
+        return allowed
+
+    def _node_mask(self, base_mask, node):
+        if node.allowed is None:
+            return base_mask
+        if base_mask is None:
+            return node.allowed
+        return base_mask & node.allowed
 
     def _sample_features(self, frac, prev_mask):
         f = self.qm.num_col
@@ -202,6 +241,12 @@ class HistGrower:
                 [(node.g, node.h) for node in frontier], dtype=torch.float32, device=self.device
             )
             node_feature_mask = self._sample_features(p.colsample_bynode, level_mask)
+            if any(node.allowed is not None for node in frontier):
+                f = self.qm.num_col
+                base = node_feature_mask if node_feature_mask is not None else torch.ones(
+                    f, dtype=torch.bool, device=self.device
+                )
+                node_feature_mask = torch.stack([self._node_mask(base, node) for node in frontier])
             splits = self.backend.find_splits(
                 hists,
                 parent_sums,
@@ -211,6 +256,7 @@ class HistGrower:
                 gamma=p.gamma,
                 min_child_weight=p.min_child_weight,
                 feature_mask=node_feature_mask,
+                monotone=self.monotone,
             )
             gains = splits["gain"].cpu().numpy()
             feats = splits["feature"].cpu().numpy()
@@ -241,18 +287,20 @@ class HistGrower:
                     [bool(dls[i]) for i, _n in to_split],
                 )
                 for (i, node), left_count in zip(to_split, counts):
-                    lid, rid = self._apply_split(
+                    lid, rid, lstate, rstate = self._apply_split(
                         tree, node, int(feats[i]), int(bins[i]), bool(dls[i]),
                         float(gains[i]), float(lgs[i]), float(lhs[i]),
                     )
                     mid = node.start + left_count
                     next_frontier.append(
-                        _Node(lid, 1 - parity, node.start, mid, float(lgs[i]), float(lhs[i]), depth + 1)
+                        _Node(lid, 1 - parity, node.start, mid, float(lgs[i]), float(lhs[i]),
+                              depth + 1, lstate[0], lstate[1], lstate[2])
                     )
                     next_frontier.append(
                         _Node(
                             rid, 1 - parity, mid, node.end,
                             node.g - float(lgs[i]), node.h - float(lhs[i]), depth + 1,
+                            rstate[0], rstate[1], rstate[2],
                         )
                     )
 
@@ -285,7 +333,7 @@ class HistGrower:
                     self._build_level_hists(gh, scale, [], [(node.nid, parent, sibling.nid)], node_hist)
                 else:
                     self._build_level_hists(gh, scale, [node], [], node_hist)
-            mask = self._sample_features(p.colsample_bynode, tree_mask)
+            mask = self._node_mask(self._sample_features(p.colsample_bynode, tree_mask), node)
             s = self.backend.find_splits(
                 node_hist[node.nid].unsqueeze(0),
                 torch.tensor([(node.g, node.h)], dtype=torch.float32, device=self.device),
@@ -295,6 +343,7 @@ class HistGrower:
                 gamma=p.gamma,
                 min_child_weight=p.min_child_weight,
                 feature_mask=mask,
+                monotone=self.monotone,
             )
             gain = float(s["gain"][0])
             if gain > 0.0:
@@ -333,13 +382,15 @@ class HistGrower:
                 [s["bin"]],
                 [s["default_left"]],
             )
-            lid, rid = self._apply_split(
+            lid, rid, lstate, rstate = self._apply_split(
                 tree, node, s["feature"], s["bin"], s["default_left"], s["gain"], s["left_g"], s["left_h"]
             )
             mid = node.start + counts[0]
-            lnode = _Node(lid, 1 - node.parity, node.start, mid, s["left_g"], s["left_h"], node.depth + 1)
+            lnode = _Node(lid, 1 - node.parity, node.start, mid, s["left_g"], s["left_h"],
+                          node.depth + 1, lstate[0], lstate[1], lstate[2])
             rnode = _Node(
-                rid, 1 - node.parity, mid, node.end, node.g - s["left_g"], node.h - s["left_h"], node.depth + 1
+                rid, 1 - node.parity, mid, node.end, node.g - s["left_g"], node.h - s["left_h"],
+                node.depth + 1, rstate[0], rstate[1], rstate[2]
             )
             nodes[lid] = lnode
             nodes[rid] = rnode
@@ -356,19 +407,42 @@ class HistGrower:
 
     # -- shared ---------------------------------------------------------------
     def _apply_split(self, tree, node, feature, bin_idx, default_left, gain, left_g, left_h):
+        """Apply one split: node values honor monotone bounds; returns
+        (lid, rid, left_state, right_state) where each state is
+        (lower, upper, allowed) for the child _Node."""
         p = self.p
         GR, HR = node.g - left_g, node.h - left_h
         cut_base = int(self.qm.cut_ptr[feature])
         threshold = float(self.qm.cuts[cut_base + bin_idx])
-        return tree.apply_split(
+
+        l_lo, l_hi = node.lower, node.upper
+        r_lo, r_hi = node.lower, node.upper
+        constraint = int(self.monotone[feature]) if self.monotone is not None else 0
+        wl = self._weight(left_g, left_h, node.lower, node.upper)
+        wr = self._weight(GR, HR, node.lower, node.upper)
+        if constraint != 0:
+            mid = 0.5 * (wl + wr)
+            if constraint > 0:
+                l_hi = min(l_hi, mid)
+                r_lo = max(r_lo, mid)
+            else:
+                l_lo = max(l_lo, mid)
+                r_hi = min(r_hi, mid)
+            wl = self._weight(left_g, left_h, l_lo, l_hi)
+            wr = self._weight(GR, HR, r_lo, r_hi)
+
+        child_allowed = self._interaction_allowed(node.allowed, feature)
+
+        lid, rid = tree.apply_split(
             node.nid,
             feature,
             threshold,
             bin_idx,
             default_left,
             gain,
-            left_value=self._weight(left_g, left_h) * p.eta,
-            right_value=self._weight(GR, HR) * p.eta,
+            left_value=wl * p.eta,
+            right_value=wr * p.eta,
             left_hess=left_h,
             right_hess=HR,
         )
+        return lid, rid, (l_lo, l_hi, child_allowed), (r_lo, r_hi, child_allowed)

@@ -56,10 +56,14 @@ def find_splits(
     gamma=0.0,
     min_child_weight=1.0,
     feature_mask=None,
+    monotone=None,
 ):
     """Best split per node from per-node histograms.
 
     hist: (k, f * stride, 2) float32; parent_sum: (k, 2) float32.
+    feature_mask: (f,) or (k, f) bool — allowed features (colsample and
+    interaction constraints). monotone: (f,) int8 in {-1, 0, 1} — splits
+    violating the implied child-weight ordering are masked out.
     Returns dict of tensors (all shape (k,)):
       feature, bin, gain, default_left, left_g, left_h
     A node with no valid split has gain <= 0.
@@ -96,6 +100,14 @@ def find_splits(
 
     parent_score = score(parent[..., 0], parent[..., 1])
 
+    def weight(g, hs):
+        ag = torch.clamp(g.abs() - alpha, min=0.0)
+        return -torch.sign(g) * ag / (hs + lam)
+
+    mono = None
+    if monotone is not None and bool((monotone != 0).any()):
+        mono = monotone.to(device).reshape(1, f, 1)
+
     results = []
     for default_left in (False, True):
         gl = scan[..., 0] + (missing[..., 0].unsqueeze(2) if default_left else 0.0)
@@ -104,6 +116,10 @@ def find_splits(
         hr = parent[..., 1] - hl
         gain = 0.5 * (score(gl, hl) + score(gr, hr) - parent_score) - gamma
         invalid = (hl < min_child_weight) | (hr < min_child_weight) | ~real_valid.unsqueeze(0)
+        if mono is not None:
+            wl = weight(gl, hl)
+            wr = weight(gr, hr)
+            invalid = invalid | ((mono > 0) & (wl > wr)) | ((mono < 0) & (wl < wr))
         gain = torch.where(invalid, torch.full_like(gain, -float("inf")), gain)
         results.append((gain, gl, hl))
 
@@ -115,7 +131,11 @@ def find_splits(
     hl = torch.where(use_left, hl_l, hl_r)
 
     if feature_mask is not None:
-        gain = torch.where(feature_mask.reshape(1, f, 1), gain, torch.full_like(gain, -float("inf")))
+        if feature_mask.dim() == 1:
+            mask = feature_mask.reshape(1, f, 1)
+        else:
+            mask = feature_mask.reshape(k, f, 1)
+        gain = torch.where(mask, gain, torch.full_like(gain, -float("inf")))
 
     flat_gain = gain.reshape(k, -1)
     best = flat_gain.argmax(dim=1)  # (k,)
