@@ -27,6 +27,8 @@ class Booster:
         self.params = dict(params or {})
         self.trees = []           # flat list of Tree
         self.tree_info = []       # class id of each tree (0 for single-output)
+        self.weight_drop = []     # per-tree scale (dart; 1.0 for gbtree)
+        self.linear_model = None  # gblinear state (models.gblinear.LinearModel)
         self.iteration_indptr = [0]
         self.num_features = num_features
         self.feature_names = feature_names
@@ -37,6 +39,10 @@ class Booster:
         self._predict_cache = None
 
     # -- basic accessors ---------------------------------------------------
+    @property
+    def booster_type(self):
+        return self.params.get("booster", "gbtree")
+
     @property
     def objective_name(self):
         return self.params.get("objective", "reg:squarederror")
@@ -83,10 +89,11 @@ class Booster:
                 self.attributes_map[k] = str(v)
 
     # -- boosting ----------------------------------------------------------
-    def add_iteration(self, trees, tree_info):
+    def add_iteration(self, trees, tree_info, weight_drop=None):
         """Append one boosting round's trees (n_outputs × num_parallel_tree)."""
         self.trees.extend(trees)
         self.tree_info.extend(tree_info)
+        self.weight_drop.extend(weight_drop if weight_drop is not None else [1.0] * len(trees))
         self.iteration_indptr.append(len(self.trees))
         self._predict_cache = None
 
@@ -103,10 +110,16 @@ class Booster:
         if iteration_range is not None and iteration_range != (0, 0):
             lo, hi = iteration_range
             hi = min(hi, self.num_boosted_rounds()) if hi else self.num_boosted_rounds()
+        if self.booster_type == "gblinear" and self.linear_model is not None:
+            margin += torch.nan_to_num(X, nan=0.0) @ self.linear_model.weights.to(device) \
+                + self.linear_model.bias.to(device)
+            return margin.squeeze(1) if k == 1 else margin
         backend = ops.backend_for(device)
         for it in range(lo, hi):
             for t in range(self.iteration_indptr[it], self.iteration_indptr[it + 1]):
-                margin[:, self.tree_info[t]] += backend.predict_tree(self.trees[t], X)
+                contrib = backend.predict_tree(self.trees[t], X)
+                scale = self.weight_drop[t] if t < len(self.weight_drop) else 1.0
+                margin[:, self.tree_info[t]] += contrib if scale == 1.0 else contrib * scale
         return margin.squeeze(1) if k == 1 else margin
 
     def predict(
@@ -249,24 +262,41 @@ class Booster:
             }
         return obj
 
+    def _gbtree_model_json(self):
+        return {
+            "gbtree_model_param": {
+                "num_trees": str(len(self.trees)),
+                "num_parallel_tree": str(self.params.get("num_parallel_tree", 1)),
+            },
+            "iteration_indptr": list(self.iteration_indptr),
+            "tree_info": list(self.tree_info),
+            "trees": [self._tree_to_json(t, i) for i, t in enumerate(self.trees)],
+        }
+
+    def _gradient_booster_json(self):
+        if self.booster_type == "gblinear":
+            flat = self.linear_model.to_flat() if self.linear_model is not None else []
+            return {
+                "model": {"weights": flat, "boosted_rounds": self.num_boosted_rounds()},
+                "name": "gblinear",
+            }
+        if self.booster_type == "dart":
+            return {
+                "model": {
+                    "gbtree": self._gbtree_model_json(),
+                    "weight_drop": [float(w) for w in self.weight_drop],
+                },
+                "name": "dart",
+            }
+        return {"model": self._gbtree_model_json(), "name": "gbtree"}
+
     def save_json(self):
         model = {
             "learner": {
                 "attributes": dict(self.attributes_map),
                 "feature_names": self.feature_names or [],
                 "feature_types": [],
-                "gradient_booster": {
-                    "model": {
-                        "gbtree_model_param": {
-                            "num_trees": str(len(self.trees)),
-                            "num_parallel_tree": str(self.params.get("num_parallel_tree", 1)),
-                        },
-                        "iteration_indptr": list(self.iteration_indptr),
-                        "tree_info": list(self.tree_info),
-                        "trees": [self._tree_to_json(t, i) for i, t in enumerate(self.trees)],
-                    },
-                    "name": "gbtree",
-                },
+                "gradient_booster": self._gradient_booster_json(),
                 "learner_model_param": {
                     "base_score": repr(self.base_score),
                     "boost_from_average": "1",
@@ -287,6 +317,7 @@ class Booster:
 
     def load_json(self, model):
         learner = model["learner"]
+        booster_name = learner["gradient_booster"].get("name", "gbtree")
         gb = learner["gradient_booster"]["model"]
         lmp = learner["learner_model_param"]
         self.params["objective"] = learner["objective"]["name"]
@@ -296,8 +327,30 @@ class Booster:
         self.num_features = int(lmp.get("num_feature", 0))
         self.feature_names = learner.get("feature_names") or None
         self.attributes_map = dict(learner.get("attributes", {}))
+        self.params["booster"] = booster_name
+
+        if booster_name == "gblinear":
+            from .gblinear import LinearModel
+
+            flat = gb.get("weights", [])
+            self.linear_model = LinearModel.from_flat(flat, self.num_features, self.n_outputs)
+            rounds = int(gb.get("boosted_rounds", 1) or 1)
+            self.trees = []
+            self.tree_info = []
+            self.weight_drop = []
+            self.iteration_indptr = list(range(rounds + 1))
+            self._objective = None
+            self._predict_cache = None
+            return self
+
+        weight_drop = None
+        if booster_name == "dart":
+            weight_drop = [float(w) for w in gb.get("weight_drop", [])]
+            gb = gb["gbtree"]
+
         self.trees = [self._tree_from_json(t) for t in gb["trees"]]
         self.tree_info = list(gb.get("tree_info", [0] * len(self.trees)))
+        self.weight_drop = weight_drop if weight_drop else [1.0] * len(self.trees)
         indptr = gb.get("iteration_indptr")
         if indptr:
             self.iteration_indptr = list(indptr)
@@ -336,6 +389,10 @@ class Booster:
         state = dict(self.__dict__)
         state["_objective"] = None
         state["_predict_cache"] = None
+        if state.get("linear_model") is not None:
+            lm = state["linear_model"]
+            lm.weights = lm.weights.cpu()
+            lm.bias = lm.bias.cpu()
         return state
 
     def __setstate__(self, state):

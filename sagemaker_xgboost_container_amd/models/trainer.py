@@ -133,7 +133,9 @@ def train(
     objective.validate_labels(y)
 
     max_bin = int(params.get("max_bin", 256))
-    qm = quantize(X, max_bin=max_bin, sample_weight=weight, comm=comm)
+    qm = None
+    if params.get("booster", "gbtree") != "gblinear":
+        qm = quantize(X, max_bin=max_bin, sample_weight=weight, comm=comm)
     backend = backend_for(device)
 
     base_margin_value = objective.base_margin(booster.base_score)
@@ -174,7 +176,30 @@ def train(
         cbs.append(EarlyStopping(rounds=early_stopping_rounds, maximize=bool(maximize)))
     container = CallbackContainer(cbs)
 
-    grower = HistGrower(qm, params, comm=comm, generator=generator)
+    booster_kind = params.get("booster", "gbtree")
+    if booster_kind == "gblinear":
+        from .gblinear import LinearModel, LinearUpdater
+
+        linear_updater = LinearUpdater(params)
+        if booster.linear_model is None:
+            booster.linear_model = LinearModel(dtrain.num_col(), n_outputs, device)
+        else:
+            booster.linear_model.weights = booster.linear_model.weights.to(device)
+            booster.linear_model.bias = booster.linear_model.bias.to(device)
+        grower = None
+    else:
+        grower = HistGrower(qm, params, comm=comm, generator=generator)
+
+    dart = None
+    if booster_kind == "dart":
+        dart = {
+            "rate_drop": float(params.get("rate_drop", 0.0)),
+            "skip_drop": float(params.get("skip_drop", 0.0)),
+            "one_drop": int(params.get("one_drop", 0)),
+            "sample_type": params.get("sample_type", "uniform"),
+            "normalize_type": params.get("normalize_type", "tree"),
+            "eta": float(params.get("eta", params.get("learning_rate", 0.3))),
+        }
 
     booster = container.before_training(booster)
 
@@ -182,20 +207,78 @@ def train(
         if container.before_iteration(booster, epoch):
             break
 
+        # -- DART: drop a subset of existing trees before computing gradients
+        dropped = []
+        drop_contrib = {}
+        if dart is not None and booster.trees:
+            if float(torch.rand((), generator=generator, device=device)) >= dart["skip_drop"]:
+                n_trees = len(booster.trees)
+                if dart["sample_type"] == "weighted":
+                    wd = torch.tensor(booster.weight_drop, device=device)
+                    probs = wd / wd.sum() * (dart["rate_drop"] * n_trees)
+                else:
+                    probs = torch.full((n_trees,), dart["rate_drop"], device=device)
+                mask = torch.rand(n_trees, generator=generator, device=device) < probs
+                dropped = mask.nonzero().flatten().tolist()
+                if not dropped and dart["one_drop"]:
+                    dropped = [int(torch.randint(n_trees, (1,), generator=generator, device=device))]
+            for t in dropped:
+                contrib = backend.predict_tree(booster.trees[t], X) * booster.weight_drop[t]
+                margin[:, booster.tree_info[t]] -= contrib
+                drop_contrib[t] = contrib
+
         gh = objective.gradients(margin.squeeze(1) if n_outputs == 1 else margin, y, weight)
 
-        round_trees = []
-        round_info = []
-        for _parallel in range(num_parallel_tree):
+        if booster_kind == "gblinear":
             for cls in range(n_outputs):
                 gh_cls = gh if n_outputs == 1 else gh[:, cls, :].contiguous()
-                tree, leaf_jobs = grower.grow(gh_cls)
-                backend.update_margins(margin[:, cls], grower.bufs, leaf_jobs)
-                round_trees.append(tree)
-                round_info.append(cls)
+                delta = linear_updater.update_round(booster.linear_model, X, gh_cls, cls)
+                margin[:, cls] += delta
+            booster.add_iteration([], [])
+            for es in eval_sets:
+                es.margin = (
+                    torch.nan_to_num(es.X, nan=0.0) @ booster.linear_model.weights
+                    + booster.linear_model.bias
+                    + float(base_margin_value)
+                )
+        else:
+            # dart scaling factors for this round's new trees
+            k_drop = len(dropped)
+            if dart is not None and k_drop > 0:
+                lr = dart["eta"]
+                if dart["normalize_type"] == "forest":
+                    new_tree_scale = 1.0 / (1.0 + lr)
+                    dropped_factor = 1.0 / (1.0 + lr)
+                else:
+                    new_tree_scale = 1.0 / (k_drop + lr)
+                    dropped_factor = k_drop / (k_drop + lr)
+            else:
+                new_tree_scale = 1.0
+                dropped_factor = 1.0
+
+            round_trees = []
+            round_info = []
+            for _parallel in range(num_parallel_tree):
+                for cls in range(n_outputs):
+                    gh_cls = gh if n_outputs == 1 else gh[:, cls, :].contiguous()
+                    tree, leaf_jobs = grower.grow(gh_cls)
+                    if new_tree_scale != 1.0:
+                        leaf_jobs = [(p_, s_, e_, v_ * new_tree_scale) for p_, s_, e_, v_ in leaf_jobs]
+                    backend.update_margins(margin[:, cls], grower.bufs, leaf_jobs)
+                    round_trees.append(tree)
+                    round_info.append(cls)
+                    for es in eval_sets:
+                        es.margin[:, cls] += backend.predict_tree(tree, es.X) * new_tree_scale
+            # rescale dropped trees and restore their (scaled) contribution
+            for t in dropped:
+                booster.weight_drop[t] *= dropped_factor
+                margin[:, booster.tree_info[t]] += drop_contrib[t] * dropped_factor
+            booster.add_iteration(round_trees, round_info, weight_drop=[new_tree_scale] * len(round_trees))
+            if dropped:
+                # dropped-tree rescaling invalidates the incremental eval
+                # margins: recompute them from the booster exactly
                 for es in eval_sets:
-                    es.margin[:, cls] += backend.predict_tree(tree, es.X)
-        booster.add_iteration(round_trees, round_info)
+                    es.margin = booster._margin(es.X).reshape(es.X.shape[0], -1).clone()
 
         # -- evaluation ----------------------------------------------------
         results = []
