@@ -72,23 +72,29 @@ def _parse_libsvm_line(line):
     head = parts[0].split(":")
     label = float(head[0])
     weight = float(head[1]) if len(head) == 2 else None
+    qid = None
     indices = []
     values = []
     for tok in parts[1:]:
         idx, _, val = tok.partition(":")
+        if idx == "qid":
+            qid = int(val)
+            continue
         indices.append(int(idx))
         values.append(float(val))
-    return label, weight, indices, values
+    return label, weight, qid, indices, values
 
 
 def _parse_libsvm_files(files):
     """Parse libsvm files (with optional <label>:<weight> extension) to CSR."""
     labels = []
     weights = []
+    qids = []
     data = []
     indices = []
     indptr = [0]
     any_weight = False
+    any_qid = False
     for path in files:
         with open(path, "r", errors="ignore") as f:
             for line in f:
@@ -98,11 +104,14 @@ def _parse_libsvm_files(files):
                 parsed = _parse_libsvm_line(line)
                 if parsed is None:
                     continue
-                label, weight, idx, val = parsed
+                label, weight, qid, idx, val = parsed
                 labels.append(label)
                 if weight is not None:
                     any_weight = True
+                if qid is not None:
+                    any_qid = True
                 weights.append(weight if weight is not None else 1.0)
+                qids.append(qid if qid is not None else 0)
                 indices.extend(idx)
                 data.extend(val)
                 indptr.append(len(indices))
@@ -115,6 +124,7 @@ def _parse_libsvm_files(files):
         csr,
         np.asarray(labels, dtype=np.float32),
         np.asarray(weights, dtype=np.float32) if any_weight else None,
+        np.asarray(qids, dtype=np.int64) if any_qid else None,
     )
 
 
@@ -139,7 +149,7 @@ def _parse_recordio_files(files):
 
 
 def _load_uri(uri):
-    """Load 'path?format=...&k=v' into (features, label, weight)."""
+    """Load 'path?format=...&k=v' into (features, label, weight[, qid])."""
     path, _, query = uri.partition("?")
     params = {k: v[0] for k, v in parse_qs(query).items()}
     fmt = params.get("format")
@@ -152,7 +162,8 @@ def _load_uri(uri):
             weight_column=int(params["weight_column"]) if "weight_column" in params else None,
         )
     if fmt == "libsvm" or fmt is None:
-        return _parse_libsvm_files(files)
+        csr, labels, weights, qids = _parse_libsvm_files(files)
+        return csr, labels, weights, qids
     if fmt == "parquet":
         return _parse_parquet_files(files)
     if fmt == "recordio-protobuf":
@@ -178,8 +189,15 @@ class DMatrix:
         self._dense = None
         self._csr = None
 
+        self._group = None
         if isinstance(data, str):
-            features, file_label, file_weight = _load_uri(data)
+            loaded = _load_uri(data)
+            if len(loaded) == 4:
+                features, file_label, file_weight, file_qid = loaded
+                if file_qid is not None:
+                    self.set_qid(file_qid)
+            else:
+                features, file_label, file_weight = loaded
             if label is None:
                 label = file_label
             if weight is None:
@@ -238,6 +256,23 @@ class DMatrix:
     def set_base_margin(self, margin):
         self._base_margin = None if margin is None else np.asarray(margin, dtype=np.float32)
 
+    def set_group(self, group):
+        """Per-query group sizes (ranking objectives)."""
+        self._group = None if group is None else np.asarray(group, dtype=np.int64)
+
+    def get_group(self):
+        return self._group
+
+    def set_qid(self, qid):
+        """Set groups from per-row query ids (consecutive runs)."""
+        qid = np.asarray(qid)
+        if qid.size == 0:
+            self._group = None
+            return
+        change = np.nonzero(np.diff(qid))[0]
+        bounds = np.concatenate([[0], change + 1, [len(qid)]])
+        self._group = np.diff(bounds).astype(np.int64)
+
     def get_float_info(self, name):
         if name == "label":
             return self.get_label()
@@ -245,7 +280,7 @@ class DMatrix:
             return self.get_weight()
         if name == "base_margin":
             return self._base_margin
-        raise exc.AlgorithmError(f"Unknown float info field: {name}")
+        return getattr(self, "_info", {}).get(name)
 
     def set_float_info(self, name, value):
         if name == "label":
@@ -255,7 +290,9 @@ class DMatrix:
         elif name == "base_margin":
             self.set_base_margin(value)
         else:
-            raise exc.AlgorithmError(f"Unknown float info field: {name}")
+            if not hasattr(self, "_info"):
+                self._info = {}
+            self._info[name] = np.asarray(value, dtype=np.float32).reshape(-1)
 
     # -- views ------------------------------------------------------------
     def to_dense(self):

@@ -149,6 +149,68 @@ def mphe(pred, y, weight=None, slope=1.0):
     return float((w * loss).sum() / w.sum())
 
 
+def _group_bounds(objective, n, device):
+    ptr = getattr(objective, "group_ptr", None)
+    if ptr is not None and int(ptr[-1]) == n:
+        p = ptr.tolist()
+        return list(zip(p[:-1], p[1:]))
+    return [(0, n)]
+
+
+def ndcg(score, y, objective=None, k=None):
+    """Mean NDCG(@k) over query groups (exponential gain)."""
+    total = 0.0
+    bounds = _group_bounds(objective, score.shape[0], score.device) if objective else [(0, score.shape[0])]
+    for a, b in bounds:
+        s, rel = score[a:b], y[a:b]
+        m = b - a
+        top = m if k is None else min(k, m)
+        order = torch.argsort(s, descending=True)[:top]
+        disc = 1.0 / torch.log2(torch.arange(top, device=s.device).float() + 2.0)
+        dcg = ((torch.pow(2.0, rel[order]) - 1.0) * disc).sum()
+        ideal = torch.sort(rel, descending=True).values[:top]
+        idcg = ((torch.pow(2.0, ideal) - 1.0) * disc).sum()
+        total += float(dcg / idcg) if float(idcg) > 0 else 1.0
+    return total / len(bounds)
+
+
+def mean_ap(score, y, objective=None, k=None):
+    """Mean average precision over query groups (binary relevance)."""
+    total = 0.0
+    bounds = _group_bounds(objective, score.shape[0], score.device) if objective else [(0, score.shape[0])]
+    for a, b in bounds:
+        s, rel = score[a:b], (y[a:b] > 0).float()
+        m = b - a
+        top = m if k is None else min(k, m)
+        order = torch.argsort(s, descending=True)[:top]
+        hits = rel[order]
+        csum = torch.cumsum(hits, 0)
+        precision_at = csum / torch.arange(1, top + 1, device=s.device).float()
+        n_pos = float(rel.sum())
+        ap = float((precision_at * hits).sum() / max(n_pos, 1.0)) if n_pos else 1.0
+        total += ap
+    return total / len(bounds)
+
+
+def cox_nloglik(margin, y):
+    """Negative Cox partial log likelihood (Breslow), mean over events."""
+    t = y.abs()
+    event = (y > 0).to(torch.float64)
+    order = torch.argsort(t)
+    m_s = margin[order].to(torch.float64)
+    e_s = event[order]
+    exp_m = torch.exp(m_s)
+    risk = torch.flip(torch.cumsum(torch.flip(exp_m, [0]), 0), [0]).clamp(min=1e-16)
+    ll = (e_s * (m_s - torch.log(risk))).sum()
+    n_events = e_s.sum().clamp(min=1.0)
+    return float(-ll / n_events)
+
+
+def interval_regression_accuracy(pred, lower, upper):
+    ok = (pred >= lower) & ((~torch.isfinite(upper)) | (pred <= upper))
+    return float(ok.float().mean())
+
+
 METRIC_NEEDS_PROB = {"logloss", "error", "auc", "aucpr", "merror", "mlogloss"}
 
 
@@ -211,4 +273,27 @@ def evaluate(metric_name, margin, y, weight, objective):
         return gamma_deviance(pred, y, weight)
     if name == "tweedie-nloglik":
         return tweedie_nloglik(pred, y, weight, rho=float(objective.params.get("tweedie_variance_power", 1.5)))
+    if name == "ndcg":
+        return ndcg(margin, y, objective, k=int(threshold) if threshold else None)
+    if name == "map":
+        return mean_ap(margin, y, objective, k=int(threshold) if threshold else None)
+    if name == "cox-nloglik":
+        return cox_nloglik(margin, y)
+    if name == "aft-nloglik":
+        lower = getattr(objective, "lower", None)
+        if lower is None:
+            lower = y.to(torch.float64)
+            upper = y.to(torch.float64)
+        else:
+            upper = objective.upper if objective.upper is not None else torch.full_like(lower, float("inf"))
+        loss = objective.nloglik(margin.to(torch.float64), lower, upper)
+        return float(loss.mean())
+    if name == "interval-regression-accuracy":
+        lower = getattr(objective, "lower", None)
+        if lower is None:
+            lower = y.to(torch.float64)
+            upper = y.to(torch.float64)
+        else:
+            upper = objective.upper if objective.upper is not None else torch.full_like(lower, float("inf"))
+        return interval_regression_accuracy(torch.exp(margin.to(torch.float64)), lower, upper)
     raise NotImplementedError(f"Eval metric '{metric_name}' is not implemented yet")

@@ -31,6 +31,9 @@ class Objective:
     def validate_labels(self, y):
         pass
 
+    def set_info(self, dmatrix, device):
+        """Pull auxiliary training info (groups, label bounds) off the DMatrix."""
+
     def base_margin(self, base_score):
         """Transform base_score (prob/mean space) into raw margin space."""
         return float(base_score)
@@ -283,6 +286,206 @@ class SquaredErrorLegacy(SquaredError):
     name = "reg:linear"
 
 
+class _RankObjective(Objective):
+    """LambdaMART-style pairwise ranking base.
+
+    Groups come from the DMatrix (qid / set_group); absent group info treats
+    the whole dataset as one query (xgboost behavior)."""
+
+    default_metric = "map"
+
+    def __init__(self, params=None):
+        super().__init__(params)
+        self.group_ptr = None  # (G+1,) int64 boundaries
+
+    def set_info(self, dmatrix, device):
+        group = dmatrix.get_group() if hasattr(dmatrix, "get_group") else None
+        if group is not None:
+            import numpy as np
+
+            ptr = np.concatenate([[0], np.cumsum(group)])
+            self.group_ptr = torch.as_tensor(ptr, dtype=torch.long, device=device)
+        else:
+            self.group_ptr = None
+
+    def _pair_weight(self, y, margin, i, j, g_start, g_end):
+        return torch.ones_like(y[i])
+
+    def gradients(self, margin, y, weight=None):
+        n = margin.shape[0]
+        device = margin.device
+        grad = torch.zeros_like(margin)
+        hess = torch.zeros_like(margin)
+        if self.group_ptr is None:
+            bounds = [(0, n)]
+        else:
+            ptr = self.group_ptr.tolist()
+            bounds = list(zip(ptr[:-1], ptr[1:]))
+        for g_start, g_end in bounds:
+            m = g_end - g_start
+            if m < 2:
+                continue
+            # sample one partner per doc (num_pairsample=1 default)
+            perm = torch.randperm(m, device=device) + g_start
+            base = torch.arange(g_start, g_end, device=device)
+            yi, yj = y[base], y[perm]
+            keep = yi != yj
+            if not bool(keep.any()):
+                continue
+            i = torch.where(yi > yj, base, perm)[keep]
+            j = torch.where(yi > yj, perm, base)[keep]
+            s = margin[i] - margin[j]
+            rho = torch.sigmoid(-s)  # 1/(1+e^s)
+            w = self._pair_weight(y, margin, i, j, g_start, g_end)
+            grad.index_add_(0, i, -rho * w)
+            grad.index_add_(0, j, rho * w)
+            h = torch.clamp(rho * (1 - rho) * w, min=1e-16)
+            hess.index_add_(0, i, h)
+            hess.index_add_(0, j, h)
+        return _pack(grad, torch.clamp(hess, min=1e-16), weight)
+
+
+class RankPairwise(_RankObjective):
+    name = "rank:pairwise"
+
+
+class RankNDCG(_RankObjective):
+    name = "rank:ndcg"
+    default_metric = "ndcg"
+
+    def _pair_weight(self, y, margin, i, j, g_start, g_end):
+        # |ΔNDCG| of swapping i and j at current ranking
+        seg = slice(g_start, g_end)
+        order = torch.argsort(margin[seg], descending=True)
+        rank = torch.empty_like(order)
+        rank[order] = torch.arange(order.numel(), device=margin.device)
+        disc = 1.0 / torch.log2(rank.float() + 2.0)
+        gain = torch.pow(2.0, y[seg]) - 1.0
+        idcg = (torch.sort(gain, descending=True).values * 1.0 /
+                torch.log2(torch.arange(gain.numel(), device=margin.device).float() + 2.0)).sum()
+        idcg = torch.clamp(idcg, min=1e-16)
+        gi, gj = torch.pow(2.0, y[i]) - 1.0, torch.pow(2.0, y[j]) - 1.0
+        di, dj = disc[i - g_start], disc[j - g_start]
+        return torch.abs((gi - gj) * (di - dj)) / idcg
+
+
+class RankMAP(_RankObjective):
+    name = "rank:map"
+    default_metric = "map"
+
+
+class SurvivalCox(Objective):
+    """Cox proportional hazards (Breslow); label > 0 = event time,
+    label < 0 = right-censored at |label|."""
+
+    name = "survival:cox"
+    default_metric = "cox-nloglik"
+
+    def base_margin(self, base_score):
+        return 0.0
+
+    def transform(self, margin):
+        return torch.exp(margin)
+
+    def gradients(self, margin, y, weight=None):
+        t = y.abs()
+        event = (y > 0).to(margin.dtype)
+        order = torch.argsort(t)  # ascending time
+        m_s = margin[order].to(torch.float64)
+        e_s = event[order].to(torch.float64)
+        exp_m = torch.exp(m_s)
+        # risk set sum for each i: sum of exp(m_j) with t_j >= t_i
+        risk = torch.flip(torch.cumsum(torch.flip(exp_m, [0]), 0), [0])
+        risk = torch.clamp(risk, min=1e-16)
+        s1 = torch.cumsum(e_s / risk, 0)
+        s2 = torch.cumsum(e_s / (risk * risk), 0)
+        grad_s = -e_s + exp_m * s1
+        hess_s = exp_m * s1 - exp_m * exp_m * s2
+        grad = torch.empty_like(grad_s)
+        hess = torch.empty_like(hess_s)
+        grad[order] = grad_s
+        hess[order] = hess_s
+        return _pack(grad.to(torch.float32), torch.clamp(hess, min=1e-16).to(torch.float32), weight)
+
+
+class SurvivalAFT(Objective):
+    """Accelerated failure time with interval censoring.
+
+    Label bounds via DMatrix.set_float_info('label_lower_bound'/'label_upper_bound');
+    a plain label means uncensored. Gradients/diag-hessians via autograd on
+    the exact per-row negative log likelihood.
+    """
+
+    name = "survival:aft"
+    default_metric = "aft-nloglik"
+
+    def __init__(self, params=None):
+        super().__init__(params)
+        self.dist = self.params.get("aft_loss_distribution", "normal")
+        self.sigma = float(self.params.get("aft_loss_distribution_scale", 1.0))
+        self.lower = None
+        self.upper = None
+
+    def set_info(self, dmatrix, device):
+        lo = dmatrix.get_float_info("label_lower_bound") if hasattr(dmatrix, "get_float_info") else None
+        hi = dmatrix.get_float_info("label_upper_bound") if hasattr(dmatrix, "get_float_info") else None
+        self.lower = None if lo is None else torch.as_tensor(lo, dtype=torch.float64, device=device)
+        self.upper = None if hi is None else torch.as_tensor(hi, dtype=torch.float64, device=device)
+
+    def base_margin(self, base_score):
+        import math
+
+        return math.log(max(base_score, 1e-16))
+
+    def transform(self, margin):
+        return torch.exp(margin)
+
+    def _log_cdf(self, z):
+        if self.dist == "normal":
+            return torch.log(torch.clamp(0.5 * (1 + torch.erf(z / 1.4142135623730951)), min=1e-300))
+        if self.dist == "logistic":
+            return torch.nn.functional.logsigmoid(z)
+        # extreme (Gumbel minimum): F(z) = 1 - exp(-exp(z))
+        return torch.log(torch.clamp(1.0 - torch.exp(-torch.exp(z)), min=1e-300))
+
+    def _log_pdf(self, z):
+        if self.dist == "normal":
+            return -0.5 * z * z - 0.9189385332046727
+        if self.dist == "logistic":
+            return z - 2.0 * torch.nn.functional.softplus(z)
+        return z - torch.exp(z)
+
+    def nloglik(self, margin, lower, upper):
+        sigma = self.sigma
+        uncensored = torch.isfinite(upper) & (lower == upper)
+        z_lo = (torch.log(torch.clamp(lower, min=1e-16)) - margin) / sigma
+        loss_unc = -(self._log_pdf(z_lo) - torch.log(torch.tensor(sigma, dtype=margin.dtype, device=margin.device)))
+        z_hi = torch.where(
+            torch.isfinite(upper), (torch.log(torch.clamp(upper, min=1e-16)) - margin) / sigma,
+            torch.full_like(margin, 50.0),
+        )
+        cdf_hi = torch.exp(self._log_cdf(z_hi))
+        cdf_lo = torch.where(lower > 0, torch.exp(self._log_cdf(z_lo)), torch.zeros_like(margin))
+        loss_cen = -torch.log(torch.clamp(cdf_hi - cdf_lo, min=1e-12))
+        return torch.where(uncensored, loss_unc, loss_cen)
+
+    def gradients(self, margin, y, weight=None):
+        device = margin.device
+        if self.lower is not None:
+            lower = self.lower
+            upper = self.upper if self.upper is not None else torch.full_like(lower, float("inf"))
+        else:
+            lower = y.to(torch.float64)
+            upper = y.to(torch.float64)
+        m = margin.detach().to(torch.float64).requires_grad_(True)
+        loss = self.nloglik(m, lower, upper).sum()
+        (g,) = torch.autograd.grad(loss, m, create_graph=True)
+        (h,) = torch.autograd.grad(g.sum(), m)
+        g = g.detach().to(torch.float32)
+        h = torch.clamp(h.detach(), min=1e-6).to(torch.float32)
+        return _pack(g, h, weight)
+
+
 OBJECTIVES = {
     cls.name: cls
     for cls in (
@@ -300,6 +503,11 @@ OBJECTIVES = {
         AbsoluteError,
         Softmax,
         SoftmaxLabel,
+        RankPairwise,
+        RankNDCG,
+        RankMAP,
+        SurvivalCox,
+        SurvivalAFT,
     )
 }
 

@@ -131,6 +131,7 @@ def train(
         else None
     )
     objective.validate_labels(y)
+    objective.set_info(dtrain, device)
 
     max_bin = int(params.get("max_bin", 256))
     qm = None
