@@ -258,12 +258,13 @@ class HistGrower:
                 feature_mask=node_feature_mask,
                 monotone=self.monotone,
             )
-            gains = splits["gain"].cpu().numpy()
-            feats = splits["feature"].cpu().numpy()
-            bins = splits["bin"].cpu().numpy()
-            dls = splits["default_left"].cpu().numpy()
-            lgs = splits["left_g"].cpu().numpy()
-            lhs = splits["left_h"].cpu().numpy()
+            packed = splits["packed"].cpu().numpy()  # one D2H transfer
+            gains = packed[:, 0]
+            feats = packed[:, 1].astype(int)
+            bins = packed[:, 2].astype(int)
+            dls = packed[:, 3] > 0.5
+            lgs = packed[:, 4]
+            lhs = packed[:, 5]
 
             # 3. decide splits (leaf cap) and batch-partition
             to_split = []
@@ -345,7 +346,8 @@ class HistGrower:
                 feature_mask=mask,
                 monotone=self.monotone,
             )
-            gain = float(s["gain"][0])
+            row = s["packed"][0].cpu().numpy()
+            gain = float(row[0])
             if gain > 0.0:
                 heapq.heappush(
                     heap,
@@ -354,11 +356,11 @@ class HistGrower:
                         counter,
                         node.nid,
                         {
-                            "feature": int(s["feature"][0]),
-                            "bin": int(s["bin"][0]),
-                            "default_left": bool(s["default_left"][0]),
-                            "left_g": float(s["left_g"][0]),
-                            "left_h": float(s["left_h"][0]),
+                            "feature": int(row[1]),
+                            "bin": int(row[2]),
+                            "default_left": bool(row[3] > 0.5),
+                            "left_g": float(row[4]),
+                            "left_h": float(row[5]),
                             "gain": gain,
                         },
                     ),

@@ -105,9 +105,51 @@ def hist_to_float(acc, scale):
     return out
 
 
-# split scan operates on (nodes x total_bins) tensors — small; the torch
-# implementation runs as a handful of fused kernels on-device.
-find_splits = torch_ref.find_splits
+def find_splits(
+    hist,
+    parent_sum,
+    qm,
+    reg_lambda=1.0,
+    reg_alpha=0.0,
+    gamma=0.0,
+    min_child_weight=1.0,
+    feature_mask=None,
+    monotone=None,
+):
+    """Fused HIP split scan: 2 kernel launches per level, zero host syncs.
+
+    Falls back to the torch reference when a feature has > 256 real bins
+    (max_bin > 256 configurations)."""
+    k = hist.shape[0]
+    f = qm.num_col
+    stride = qm.stride
+    if not hasattr(qm, "_nbins_i32"):
+        qm._nbins_i32 = qm.nbins.to(torch.int32).contiguous()
+    if int(qm._nbins_i32.max()) > 256:
+        return torch_ref.find_splits(
+            hist, parent_sum, qm, reg_lambda=reg_lambda, reg_alpha=reg_alpha, gamma=gamma,
+            min_child_weight=min_child_weight, feature_mask=feature_mask, monotone=monotone,
+        )
+    device = hist.device
+    if feature_mask is None:
+        mask = torch.empty(0, dtype=torch.uint8, device=device)
+        per_node = 0
+    else:
+        mask = feature_mask.to(torch.uint8).contiguous()
+        per_node = 1 if feature_mask.dim() == 2 else 0
+    mono = (
+        monotone.to(torch.int8).contiguous()
+        if monotone is not None
+        else torch.empty(0, dtype=torch.int8, device=device)
+    )
+    cands = torch.empty((k, f, 5), dtype=torch.float32, device=device)
+    out = torch.empty((k, 6), dtype=torch.float32, device=device)
+    _K.find_splits(
+        hist.contiguous(), parent_sum.contiguous(), qm._nbins_i32, mask, mono, cands, out,
+        k, f, stride, 1 if qm.has_missing else 0, per_node,
+        reg_lambda, reg_alpha, gamma, min_child_weight,
+    )
+    return {"packed": out}
 
 
 def partition_level(qm, src, dst, segs, feats, split_bins, default_lefts):

@@ -148,13 +148,26 @@ def find_splits(
     out_gl = gl.reshape(k, -1).gather(1, idx).squeeze(1)
     out_hl = hl.reshape(k, -1).gather(1, idx).squeeze(1)
 
+    gain_out = torch.where(torch.isinf(best_gain), torch.full_like(best_gain, -1.0), best_gain).to(torch.float32)
+    packed = torch.stack(
+        [
+            gain_out,
+            best_feat.to(torch.float32),
+            best_bin.to(torch.float32),
+            out_default_left.to(torch.float32),
+            out_gl.to(torch.float32),
+            out_hl.to(torch.float32),
+        ],
+        dim=1,
+    )
     return {
         "feature": best_feat.to(torch.int32),
         "bin": best_bin.to(torch.int32),
-        "gain": torch.where(torch.isinf(best_gain), torch.full_like(best_gain, -1.0), best_gain).to(torch.float32),
+        "gain": gain_out,
         "default_left": out_default_left,
         "left_g": out_gl.to(torch.float32),
         "left_h": out_hl.to(torch.float32),
+        "packed": packed,
     }
 
 
