@@ -77,22 +77,36 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_kernel(
   __syncthreads();
 
   const int chunk = blockIdx.x - job.first_block;
-  // Feature-per-lane layout: lane l of each 32-lane half-wave owns feature
-  // (l & 31); a wave covers 2 rows x 32 features. Lanes of one wave then
-  // target DIFFERENT features' LDS regions, so same-address LDS-atomic
-  // serialization within a wave disappears (row-per-thread layout had all
-  // 64 lanes hammering the same feature's hot bins in lockstep).
-  const int lane_feat = threadIdx.x & 31;
-  const int row_slot = threadIdx.x >> 5;           // 0..(blockDim/32 - 1)
-  const int rows_per_iter = blockDim.x >> 5;       // 8 rows per 256-thread block
-  const long long step = (long long)job.num_blocks * rows_per_iter;
-  for (long long r = job.start + (long long)chunk * rows_per_iter + row_slot; r < job.end; r += step) {
+  const long long step = (long long)job.num_blocks * blockDim.x;
+  for (long long r = job.start + (long long)chunk * blockDim.x + threadIdx.x; r < job.end; r += step) {
     const int row = rowbuf[r];
     const float2 gp = gh[row];
     const unsigned long long gfix = (unsigned long long)(long long)llrintf(gp.x * scale_g);
     const unsigned long long hfix = (unsigned long long)(long long)llrintf(gp.y * scale_h);
     const BinT* rp = bins + (long long)row * nfeat + job.fg_start;
-    for (int f = lane_feat; f < nf_group; f += 32) {
+    if constexpr (sizeof(BinT) == 1) {
+      // vectorized path: 4 bins per dword load (valid when the group is
+      // 4-aligned in the row — guaranteed by host packing for nfeat%4==0)
+      if ((nf_group & 3) == 0 && ((((long long)row * nfeat + job.fg_start) & 3) == 0)) {
+        const uchar4* rp4 = reinterpret_cast<const uchar4*>(rp);
+        #pragma unroll 2
+        for (int f4 = 0; f4 < (nf_group >> 2); ++f4) {
+          const uchar4 b4 = rp4[f4];
+          const int base = (f4 << 2) * stride;
+          atomicAdd(&lhist[(base + (int)b4.x) * 2], gfix);
+          atomicAdd(&lhist[(base + (int)b4.x) * 2 + 1], hfix);
+          atomicAdd(&lhist[(base + stride + (int)b4.y) * 2], gfix);
+          atomicAdd(&lhist[(base + stride + (int)b4.y) * 2 + 1], hfix);
+          atomicAdd(&lhist[(base + 2 * stride + (int)b4.z) * 2], gfix);
+          atomicAdd(&lhist[(base + 2 * stride + (int)b4.z) * 2 + 1], hfix);
+          atomicAdd(&lhist[(base + 3 * stride + (int)b4.w) * 2], gfix);
+          atomicAdd(&lhist[(base + 3 * stride + (int)b4.w) * 2 + 1], hfix);
+        }
+        continue;
+      }
+    }
+    #pragma unroll 4
+    for (int f = 0; f < nf_group; ++f) {
       const int slot = (f * stride + (int)rp[f]) * 2;
       atomicAdd(&lhist[slot], gfix);
       atomicAdd(&lhist[slot + 1], hfix);
