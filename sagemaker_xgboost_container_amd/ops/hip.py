@@ -28,10 +28,13 @@ except ImportError as _e:  # pragma: no cover - exercised only on GPU boxes
 # rows stay inside int64 with 2^29 headroom.
 _FIXED_BITS = 33.0
 
-# LDS histogram slab per feature group (56 KiB -> 2 blocks/CU co-residency)
-_LDS_BYTES = 56 * 1024
-_ROWS_PER_BLOCK = 8192
-_MAX_BLOCKS_PER_JOB = 512
+import os as _os
+
+# LDS histogram slab per feature group (56 KiB -> 2 blocks/CU co-residency).
+# Env-tunable for measured sweeps (SMXGB_LDS_KB / SMXGB_ROWS_PER_BLOCK).
+_LDS_BYTES = int(_os.environ.get("SMXGB_LDS_KB", "56")) * 1024
+_ROWS_PER_BLOCK = int(_os.environ.get("SMXGB_ROWS_PER_BLOCK", "8192"))
+_MAX_BLOCKS_PER_JOB = int(_os.environ.get("SMXGB_MAX_BLOCKS", "512"))
 
 
 def compute_scale(gh, comm=None):
