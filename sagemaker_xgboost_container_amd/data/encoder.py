@@ -27,7 +27,7 @@ def csv_to_dmatrix(input, dtype=None):
     csv_string = input.decode() if isinstance(input, bytes) else input
     sniff_delimiter = csv_module.Sniffer().sniff(csv_string.split("\n")[0][:512]).delimiter
     delimiter = "," if sniff_delimiter.isalnum() else sniff_delimiter
-    logging.info("Determined delimiter of CSV input is '%s'", delimiter)
+    logging.debug("Determined delimiter of CSV input is '%s'", delimiter)
     rows = [_clean_csv_string(line, delimiter) for line in csv_string.split("\n") if line.strip()]
     np_payload = np.array(rows).astype(dtype or np.float32)
     return DMatrix(np_payload)

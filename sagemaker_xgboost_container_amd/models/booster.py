@@ -114,12 +114,18 @@ class Booster:
             margin += torch.nan_to_num(X, nan=0.0) @ self.linear_model.weights.to(device) \
                 + self.linear_model.bias.to(device)
             return margin.squeeze(1) if k == 1 else margin
-        backend = ops.backend_for(device)
-        for it in range(lo, hi):
-            for t in range(self.iteration_indptr[it], self.iteration_indptr[it + 1]):
-                contrib = backend.predict_tree(self.trees[t], X)
-                scale = self.weight_drop[t] if t < len(self.weight_drop) else 1.0
-                margin[:, self.tree_info[t]] += contrib if scale == 1.0 else contrib * scale
+        if self.trees:
+            backend = ops.backend_for(device)
+            # cached flat forest: ONE traversal kernel for all trees
+            key = str(device)
+            cache = self._predict_cache or {}
+            if key not in cache:
+                wd = self.weight_drop if any(w != 1.0 for w in self.weight_drop) else None
+                cache[key] = backend.make_flat_forest(self.trees, self.tree_info, wd, device)
+                self._predict_cache = cache
+            t_begin = self.iteration_indptr[lo]
+            t_end = self.iteration_indptr[hi]
+            margin += backend.predict_forest_flat(cache[key], X, k, t_begin, t_end)
         return margin.squeeze(1) if k == 1 else margin
 
     def predict(

@@ -232,3 +232,41 @@ def predict_forest(trees, tree_info, X, k, t_begin=0, t_end=None, out=None):
 def predict_tree(tree, X):
     out = predict_forest([tree], [0], X, 1)
     return out[:, 0]
+
+
+def make_flat_forest(trees, tree_info, weight_drop, device):
+    """Flatten trees for the HIP forest kernel (int32 ids, u8 default-left)."""
+    offsets = np.cumsum([0] + [t.num_nodes for t in trees]).astype(np.int32)
+    left = np.concatenate([t.left + (t.left >= 0) * offsets[i] for i, t in enumerate(trees)])
+    right = np.concatenate([t.right + (t.right >= 0) * offsets[i] for i, t in enumerate(trees)])
+    value = np.concatenate(
+        [t.value * (weight_drop[i] if weight_drop else 1.0) for i, t in enumerate(trees)]
+    ).astype(np.float32)
+    return {
+        "left": torch.from_numpy(left.astype(np.int32)).to(device),
+        "right": torch.from_numpy(right.astype(np.int32)).to(device),
+        "feature": torch.from_numpy(np.concatenate([t.feature for t in trees]).astype(np.int32)).to(device),
+        "threshold": torch.from_numpy(np.concatenate([t.threshold for t in trees]).astype(np.float32)).to(device),
+        "default_left": torch.from_numpy(
+            np.concatenate([t.default_left for t in trees]).astype(np.uint8)
+        ).to(device),
+        "value": torch.from_numpy(value).to(device),
+        "tree_root": torch.from_numpy(offsets[:-1]).to(device),
+        "tree_cls": torch.from_numpy(np.asarray(tree_info, dtype=np.int32)).to(device),
+        "n_trees": len(trees),
+    }
+
+
+def predict_forest_flat(flat, X, k, t_begin=0, t_end=None):
+    if t_end is None:
+        t_end = flat["n_trees"]
+    n = X.shape[0]
+    out = torch.zeros((n, k), dtype=torch.float32, device=X.device)
+    if t_end <= t_begin:
+        return out
+    _K.predict_forest(
+        X.contiguous(), flat["left"], flat["right"], flat["feature"], flat["threshold"],
+        flat["default_left"], flat["value"], flat["tree_root"], flat["tree_cls"],
+        t_begin, t_end, out, k,
+    )
+    return out

@@ -230,3 +230,57 @@ def predict_tree(tree, X):
         node[active] = torch.where(go_left, left[cur], right[cur])
         active = ~is_leaf[node]
     return value[node]
+
+
+def make_flat_forest(trees, tree_info, weight_drop, device):
+    """Flatten trees into device arrays with global node ids; leaf values
+    pre-scaled by weight_drop (dart)."""
+    import numpy as np
+
+    offsets = np.cumsum([0] + [t.num_nodes for t in trees]).astype(np.int64)
+    left = np.concatenate([t.left + (t.left >= 0) * offsets[i] for i, t in enumerate(trees)])
+    right = np.concatenate([t.right + (t.right >= 0) * offsets[i] for i, t in enumerate(trees)])
+    value = np.concatenate(
+        [t.value * (weight_drop[i] if weight_drop else 1.0) for i, t in enumerate(trees)]
+    ).astype(np.float32)
+    return {
+        "left": torch.from_numpy(left.astype(np.int64)).to(device),
+        "right": torch.from_numpy(right.astype(np.int64)).to(device),
+        "feature": torch.from_numpy(np.concatenate([t.feature for t in trees]).astype(np.int64)).to(device),
+        "threshold": torch.from_numpy(np.concatenate([t.threshold for t in trees]).astype(np.float32)).to(device),
+        "default_left": torch.from_numpy(np.concatenate([t.default_left for t in trees])).to(device),
+        "value": torch.from_numpy(value).to(device),
+        "tree_root": torch.from_numpy(offsets[:-1].astype(np.int64)).to(device),
+        "tree_cls": torch.from_numpy(np.asarray(tree_info, dtype=np.int64)).to(device),
+        "n_trees": len(trees),
+    }
+
+
+def predict_forest_flat(flat, X, k, t_begin=0, t_end=None):
+    """(n, k) margin contributions of trees [t_begin, t_end) — vectorized
+    level-synchronized traversal over all (row, tree) pairs."""
+    if t_end is None:
+        t_end = flat["n_trees"]
+    n = X.shape[0]
+    T = t_end - t_begin
+    out = torch.zeros((n, k), dtype=torch.float32, device=X.device)
+    if T <= 0:
+        return out
+    node = flat["tree_root"][t_begin:t_end].unsqueeze(0).expand(n, T).contiguous()
+    left = flat["left"]
+    right = flat["right"]
+    feat = flat["feature"]
+    thresh = flat["threshold"]
+    defl = flat["default_left"]
+    active = left[node] >= 0
+    while bool(active.any()):
+        cur = node[active]
+        fidx = feat[cur]
+        rows = active.nonzero(as_tuple=True)[0]
+        fv = X[rows, fidx]
+        missing = torch.isnan(fv)
+        go_left = torch.where(missing, defl[cur], fv < thresh[cur])
+        node[active] = torch.where(go_left, left[cur], right[cur])
+        active = left[node] >= 0
+    out.index_add_(1, flat["tree_cls"][t_begin:t_end], flat["value"][node])
+    return out
