@@ -1,6 +1,9 @@
 import os
 import sys
 
+# Import pyarrow before the HTTP stack (httpx/fastapi/charset_normalizer):
+# loading pyarrow.parquet after them segfaults in this image.
+import pyarrow.parquet  # noqa: F401
 import pytest
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
