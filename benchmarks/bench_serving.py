@@ -59,20 +59,40 @@ def main():
     serve.ScoringService.MODEL_PATH = model_dir
     serve.ScoringService.reset()
 
-    from fastapi.testclient import TestClient
+    # real server: uvicorn in a background thread + raw stdlib HTTP client
+    import http.client as http_client
+    import logging as _logging
+    import threading
 
-    client = TestClient(serve.ScoringService.app)
+    import uvicorn
+
+    _logging.getLogger("uvicorn").setLevel(_logging.WARNING)
+    _logging.getLogger("uvicorn.access").setLevel(_logging.WARNING)
+    port = 18080
+    config = uvicorn.Config(serve.ScoringService.app, host="127.0.0.1", port=port, log_level="warning")
+    server = uvicorn.Server(config)
+    thread = threading.Thread(target=server.run, daemon=True)
+    thread.start()
+    for _ in range(100):
+        if server.started:
+            break
+        time.sleep(0.05)
+
     X = rng.normal(size=(args.rows, args.features)).astype(np.float32)
     payload = "\n".join(",".join(f"{v:.6f}" for v in row) for row in X).encode()
 
+    conn = http_client.HTTPConnection("127.0.0.1", port)
     lat = []
     for i in range(args.warmup + args.requests):
         t0 = time.perf_counter()
-        r = client.post("/invocations", content=payload, headers={"Content-Type": "text/csv"})
-        assert r.status_code == 200, r.text
+        conn.request("POST", "/invocations", body=payload, headers={"Content-Type": "text/csv"})
+        resp = conn.getresponse()
+        body = resp.read()
+        assert resp.status == 200, body[:200]
         if i >= args.warmup:
             lat.append((time.perf_counter() - t0) * 1000)
     lat = np.array(lat)
+    server.should_exit = True
 
     # raw predict path (model loaded, DMatrix pre-parsed)
     from sagemaker_xgboost_container_amd.algorithm_mode import serve_utils
