@@ -368,12 +368,27 @@ class Booster:
         return self
 
     def load_model(self, path):
+        """Load a Booster file: JSON or UBJSON (xgboost >= 1.6 default)."""
         with open(path, "rb") as f:
-            head = f.read(1)
-        if head not in (b"{",):
-            raise ValueError(f"Unsupported model format in {path} (expected JSON Booster)")
-        with open(path, "r") as f:
-            return self.load_json(json.load(f))
+            raw = f.read()
+        head = raw[:1]
+        if head == b"{" and raw.lstrip()[:1] == b"{":
+            try:
+                return self.load_json(json.loads(raw))
+            except (json.JSONDecodeError, UnicodeDecodeError):
+                pass  # '{' is also the UBJSON object marker — fall through
+        if head == b"{":
+            from ..utils import ubjson
+
+            return self.load_json(ubjson.loads(raw))
+        raise ValueError(f"Unsupported model format in {path} (expected JSON/UBJSON Booster)")
+
+    def save_model_ubj(self, path):
+        from ..utils import ubjson
+
+        with open(str(path) + ".tmp", "wb") as f:
+            f.write(ubjson.dumps(self.save_json()))
+        os.replace(str(path) + ".tmp", str(path))
 
     def save_config(self):
         return json.dumps(
