@@ -85,8 +85,32 @@ def _parse_libsvm_line(line):
     return label, weight, qid, indices, values
 
 
+def _parse_libsvm_files_native(files):
+    """Multi-threaded C++ libsvm parser (ops/csrc/text_parsers.cpp)."""
+    import torch  # noqa: F401  (extension links against torch)
+
+    from ..ops import _smxgb_hip as K
+
+    text = "".join(open(f, "r", errors="ignore").read() for f in files)
+    values, indices, indptr, labels, weights, qids, ncol = K.parse_libsvm(text, 0)
+    csr = sp.csr_matrix(
+        (values.numpy(), indices.numpy(), indptr.numpy()),
+        shape=(len(labels), max(int(ncol[0]), 0)),
+    )
+    return (
+        csr,
+        labels.numpy(),
+        weights.numpy() if weights.numel() else None,
+        qids.numpy() if qids.numel() else None,
+    )
+
+
 def _parse_libsvm_files(files):
     """Parse libsvm files (with optional <label>:<weight> extension) to CSR."""
+    try:
+        return _parse_libsvm_files_native(files)
+    except ImportError:
+        pass  # extension not built: pure-Python fallback below
     labels = []
     weights = []
     qids = []

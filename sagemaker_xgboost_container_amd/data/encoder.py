@@ -82,11 +82,19 @@ def recordio_protobuf_to_dmatrix(string_like):
     return DMatrix(features, label=labels)
 
 
+def npy_to_dmatrix(bytes_like):
+    """Numpy .npy payload -> DMatrix. allow_pickle stays False (the
+    hardening the reference ships as docker/.../patches/decoder.py)."""
+    stream = io.BytesIO(bytes(bytes_like))
+    return DMatrix(np.load(stream, allow_pickle=False))
+
+
 _dmatrix_decoders_map = {
     _MIME_CSV: csv_to_dmatrix,
     xgb_content_types.LIBSVM: libsvm_to_dmatrix,
     xgb_content_types.X_LIBSVM: libsvm_to_dmatrix,
     xgb_content_types.X_RECORDIO_PROTOBUF: recordio_protobuf_to_dmatrix,
+    "application/x-npy": npy_to_dmatrix,
 }
 
 

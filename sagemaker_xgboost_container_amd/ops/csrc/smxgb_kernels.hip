@@ -541,7 +541,10 @@ void find_splits(torch::Tensor hist, torch::Tensor parent, torch::Tensor nbins,
                      (int)f);
 }
 
+void init_text_parsers(pybind11::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  init_text_parsers(m);
   m.def("hist_build", &hist_build, "batched LDS-staged fixed-point histogram build");
   m.def("hist_convert", &hist_convert, "fixed-point -> float32 histogram convert");
   m.def("partition", &partition, "batched two-ended row partition");

@@ -22,7 +22,10 @@ setup(
     ext_modules=[
         CUDAExtension(
             name="sagemaker_xgboost_container_amd.ops._smxgb_hip",
-            sources=["sagemaker_xgboost_container_amd/ops/csrc/smxgb_kernels.hip"],
+            sources=[
+                "sagemaker_xgboost_container_amd/ops/csrc/smxgb_kernels.hip",
+                "sagemaker_xgboost_container_amd/ops/csrc/text_parsers.cpp",
+            ],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "--offload-arch=gfx950"],
