@@ -208,7 +208,8 @@ class HistGrower:
     # -- depthwise ----------------------------------------------------------
     def _grow_depthwise(self, tree, gh, scale, root_node, tree_mask):
         p = self.p
-        node_hist = {}
+        prev_hists = None  # previous level's fused hist tensor
+        prev_row = {}      # nid -> row in prev_hists
         finished = []  # _Node leaves
         frontier = [root_node]
         n_leaves = 1
@@ -233,10 +234,35 @@ class HistGrower:
                         build_nodes.append(node)
                     else:
                         derived.append((node.nid, parent, sib_nid))
-            self._build_level_hists(gh, scale, build_nodes, derived, node_hist)
+
+            # one fused level tensor: built rows copied in, derived rows =
+            # prev-level parent minus sibling (single batched subtraction)
+            frontier_row = {node.nid: i for i, node in enumerate(frontier)}
+            slots = self.qm.total_slots
+            hists = torch.empty((len(frontier), slots, 2), dtype=torch.float32, device=self.device)
+            if build_nodes:
+                rowbuf = self._bufs[build_nodes[0].parity]
+                acc = self.backend.build_histograms(
+                    self.qm, gh, rowbuf, [n.seg for n in build_nodes], scale
+                )
+                self._allreduce(acc)
+                built = self.backend.hist_to_float(acc, scale)
+                build_rows = torch.tensor(
+                    [frontier_row[n.nid] for n in build_nodes], dtype=torch.long, device=self.device
+                )
+                hists.index_copy_(0, build_rows, built)
+            if derived:
+                drows = torch.tensor([frontier_row[nid] for nid, _p, _s in derived],
+                                     dtype=torch.long, device=self.device)
+                prows = torch.tensor([prev_row[pnid] for _n, pnid, _s in derived],
+                                     dtype=torch.long, device=self.device)
+                srows = torch.tensor([frontier_row[snid] for _n, _p, snid in derived],
+                                     dtype=torch.long, device=self.device)
+                hists.index_copy_(
+                    0, drows, prev_hists.index_select(0, prows) - hists.index_select(0, srows)
+                )
 
             # 2. batched split search
-            hists = torch.stack([node_hist[node.nid] for node in frontier])
             parent_sums = torch.tensor(
                 [(node.g, node.h) for node in frontier], dtype=torch.float32, device=self.device
             )
@@ -305,16 +331,12 @@ class HistGrower:
                         )
                     )
 
-            # free hists no longer needed (grandparents of the next frontier)
-            for node in frontier:
-                parent = int(tree.parent[node.nid])
-                if parent >= 0:
-                    node_hist.pop(parent, None)
+            prev_hists = hists
+            prev_row = frontier_row
             frontier = next_frontier
             depth += 1
 
         finished.extend(frontier)
-        node_hist.clear()
         return [(n.parity, n.start, n.end, float(tree.value[n.nid])) for n in finished]
 
     # -- lossguide ----------------------------------------------------------
