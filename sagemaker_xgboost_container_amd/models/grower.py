@@ -82,7 +82,7 @@ class HistGrower:
         self.backend = ops.backend_for(qm.device)
         self.device = qm.device
         self.generator = generator
-        self._bufs = None
+        self.state = None
         f = qm.num_col
         self.monotone = None
         if self.p.monotone_constraints:
@@ -147,9 +147,8 @@ class HistGrower:
     def _build_level_hists(self, gh, scale, build_nodes, derived, node_hist):
         """build_nodes: [_Node] (all same parity); derived: [(nid, parent, sib)]."""
         if build_nodes:
-            rowbuf = self._bufs[build_nodes[0].parity]
-            acc = self.backend.build_histograms(
-                self.qm, gh, rowbuf, [n.seg for n in build_nodes], scale
+            acc = self.state.build_histograms(
+                [n.seg for n in build_nodes], build_nodes[0].parity, scale
             )
             self._allreduce(acc)
             hist = self.backend.hist_to_float(acc, scale)
@@ -174,20 +173,19 @@ class HistGrower:
             keep = torch.rand(n, device=self.device, generator=self.generator) < p.subsample
             rows = keep.nonzero(as_tuple=True)[0].to(torch.int32)
         else:
-            rows = torch.arange(n, dtype=torch.int32, device=self.device)
+            rows = None  # full data: level 0 streams the original matrix
 
-        cap = rows.numel()
-        if self._bufs is None or self._bufs[0].numel() < cap:
-            self._bufs = (
-                torch.empty(max(cap, n), dtype=torch.int32, device=self.device),
-                torch.empty(max(cap, n), dtype=torch.int32, device=self.device),
-            )
-        self._bufs[0][:cap] = rows
+        self.state = self.backend.make_tree_state(qm, gh, rows)
+        cap = self.state.cap
 
         scale = self.backend.compute_scale(gh, comm=self.comm)
 
         tree = Tree()
-        root_sum = gh.index_select(0, rows.long()).to(torch.float64).sum(0)
+        root_sum = (
+            gh.to(torch.float64).sum(0)
+            if rows is None
+            else gh.index_select(0, rows.long()).to(torch.float64).sum(0)
+        )
         root_sum = self._allreduce(root_sum)
         G, H = float(root_sum[0]), float(root_sum[1])
         root = tree.add_node(parent=-1, value=self._weight(G, H) * p.eta, sum_hess=H)
@@ -201,9 +199,6 @@ class HistGrower:
             leaf_jobs = self._grow_depthwise(tree, gh, scale, root_node, tree_mask)
         return tree, leaf_jobs
 
-    @property
-    def bufs(self):
-        return self._bufs
 
     # -- depthwise ----------------------------------------------------------
     def _grow_depthwise(self, tree, gh, scale, root_node, tree_mask):
@@ -241,9 +236,8 @@ class HistGrower:
             slots = self.qm.total_slots
             hists = torch.empty((len(frontier), slots, 2), dtype=torch.float32, device=self.device)
             if build_nodes:
-                rowbuf = self._bufs[build_nodes[0].parity]
-                acc = self.backend.build_histograms(
-                    self.qm, gh, rowbuf, [n.seg for n in build_nodes], scale
+                acc = self.state.build_histograms(
+                    [n.seg for n in build_nodes], build_nodes[0].parity, scale
                 )
                 self._allreduce(acc)
                 built = self.backend.hist_to_float(acc, scale)
@@ -304,14 +298,12 @@ class HistGrower:
             next_frontier = []
             if to_split:
                 parity = to_split[0][1].parity
-                counts = self.backend.partition_level(
-                    self.qm,
-                    self._bufs[parity],
-                    self._bufs[1 - parity],
+                counts = self.state.partition_level(
                     [node.seg for _i, node in to_split],
                     [int(feats[i]) for i, _n in to_split],
                     [int(bins[i]) for i, _n in to_split],
                     [bool(dls[i]) for i, _n in to_split],
+                    parity,
                 )
                 for (i, node), left_count in zip(to_split, counts):
                     lid, rid, lstate, rstate = self._apply_split(
@@ -397,14 +389,8 @@ class HistGrower:
             node = nodes[nid]
             if p.max_depth and node.depth >= p.max_depth:
                 continue
-            counts = self.backend.partition_level(
-                self.qm,
-                self._bufs[node.parity],
-                self._bufs[1 - node.parity],
-                [node.seg],
-                [s["feature"]],
-                [s["bin"]],
-                [s["default_left"]],
+            counts = self.state.partition_level(
+                [node.seg], [s["feature"]], [s["bin"]], [s["default_left"]], node.parity
             )
             lid, rid, lstate, rstate = self._apply_split(
                 tree, node, s["feature"], s["bin"], s["default_left"], s["gain"], s["left_g"], s["left_h"]

@@ -265,7 +265,7 @@ def train(
                     tree, leaf_jobs = grower.grow(gh_cls)
                     if new_tree_scale != 1.0:
                         leaf_jobs = [(p_, s_, e_, v_ * new_tree_scale) for p_, s_, e_, v_ in leaf_jobs]
-                    backend.update_margins(margin[:, cls], grower.bufs, leaf_jobs)
+                    grower.state.update_margins(margin[:, cls], leaf_jobs)
                     round_trees.append(tree)
                     round_info.append(cls)
                     for es in eval_sets:

@@ -134,6 +134,152 @@ __global__ void hist_convert_kernel(const unsigned long long* __restrict__ in,
 }
 
 // ---------------------------------------------------------------------------
+// COMPACT-LAYOUT pipeline (v2)
+//
+// The gather-based hist kernel reads each scattered row through a full
+// 128-B cache line (28 useful bytes): at depth>0 the updater is bound by
+// that amplification. v2 keeps rows PHYSICALLY node-contiguous: partition
+// rewrites each surviving row's (row id, bin bytes, gradient pair) into
+// ping-pong compact SoA buffers, so every level's histogram build is a
+// perfectly coalesced stream and no kernel gathers by row id again.
+// ---------------------------------------------------------------------------
+
+// hist over compact buffers: rows are positions [start,end) in bins_c/gh_c.
+template <typename BinT>
+__global__ __launch_bounds__(HIST_BLOCK) void hist_compact_kernel(
+    const BinT* __restrict__ bins_c, const float2* __restrict__ gh_c,
+    const HistJob* __restrict__ jobs, const int* __restrict__ block_job,
+    unsigned long long* __restrict__ out, int nfeat, int stride,
+    float scale_g, float scale_h) {
+  extern __shared__ unsigned long long lhist[];
+
+  const HistJob job = jobs[block_job[blockIdx.x]];
+  const int nf_group = job.fg_end - job.fg_start;
+  const int lds_words = nf_group * stride * 2;
+  for (int i = threadIdx.x; i < lds_words; i += blockDim.x) lhist[i] = 0ull;
+  __syncthreads();
+
+  const int chunk = blockIdx.x - job.first_block;
+  const long long step = (long long)job.num_blocks * blockDim.x;
+  for (long long r = job.start + (long long)chunk * blockDim.x + threadIdx.x; r < job.end; r += step) {
+    const float2 gp = gh_c[r];
+    const unsigned long long gfix = (unsigned long long)(long long)llrintf(gp.x * scale_g);
+    const unsigned long long hfix = (unsigned long long)(long long)llrintf(gp.y * scale_h);
+    const BinT* rp = bins_c + (long long)r * nfeat + job.fg_start;
+    if constexpr (sizeof(BinT) == 1) {
+      if ((nf_group & 3) == 0 && ((((long long)r * nfeat + job.fg_start) & 3) == 0)) {
+        const uchar4* rp4 = reinterpret_cast<const uchar4*>(rp);
+        #pragma unroll 2
+        for (int f4 = 0; f4 < (nf_group >> 2); ++f4) {
+          const uchar4 b4 = rp4[f4];
+          const int base = (f4 << 2) * stride;
+          atomicAdd(&lhist[(base + (int)b4.x) * 2], gfix);
+          atomicAdd(&lhist[(base + (int)b4.x) * 2 + 1], hfix);
+          atomicAdd(&lhist[(base + stride + (int)b4.y) * 2], gfix);
+          atomicAdd(&lhist[(base + stride + (int)b4.y) * 2 + 1], hfix);
+          atomicAdd(&lhist[(base + 2 * stride + (int)b4.z) * 2], gfix);
+          atomicAdd(&lhist[(base + 2 * stride + (int)b4.z) * 2 + 1], hfix);
+          atomicAdd(&lhist[(base + 3 * stride + (int)b4.w) * 2], gfix);
+          atomicAdd(&lhist[(base + 3 * stride + (int)b4.w) * 2 + 1], hfix);
+        }
+        continue;
+      }
+    }
+    #pragma unroll 4
+    for (int f = 0; f < nf_group; ++f) {
+      const int slot = (f * stride + (int)rp[f]) * 2;
+      atomicAdd(&lhist[slot], gfix);
+      atomicAdd(&lhist[slot + 1], hfix);
+    }
+  }
+  __syncthreads();
+
+  unsigned long long* gout =
+      out + ((long long)job.hist_idx * nfeat + job.fg_start) * (long long)stride * 2;
+  for (int i = threadIdx.x; i < lds_words; i += blockDim.x) {
+    const unsigned long long v = lhist[i];
+    if (v) atomicAdd(&gout[i], v);
+  }
+}
+
+// compact partition: decide per row from src bins (sequential read), then
+// copy the whole record (row id + bin bytes + gh) to its contiguous
+// destination run. Destinations inside one tile form two coalesced runs.
+#define CPART_TILE 2048
+
+template <typename BinT>
+__global__ __launch_bounds__(HIST_BLOCK) void partition_compact_kernel(
+    const BinT* __restrict__ src_bins, const float2* __restrict__ src_gh,
+    const int* __restrict__ src_rows, BinT* __restrict__ dst_bins,
+    float2* __restrict__ dst_gh, int* __restrict__ dst_rows,
+    const PartJob* __restrict__ jobs, const int* __restrict__ block_job,
+    int* __restrict__ counters, int nfeat, int missing_bin) {
+  __shared__ int ldest[CPART_TILE];  // destination index per tile row
+  __shared__ int lcnt, rcnt, lbase, rbase;
+
+  const int j = block_job[blockIdx.x];
+  const PartJob job = jobs[j];
+  const int chunk = blockIdx.x - job.first_block;
+  const long long tile_step = (long long)job.num_blocks * CPART_TILE;
+
+  for (long long tile = job.start + (long long)chunk * CPART_TILE; tile < job.end; tile += tile_step) {
+    if (threadIdx.x == 0) {
+      lcnt = 0;
+      rcnt = 0;
+    }
+    __syncthreads();
+    const int tile_n = (int)min((long long)CPART_TILE, job.end - tile);
+    for (int i = threadIdx.x; i < tile_n; i += blockDim.x) {
+      const int b = (int)src_bins[(tile + i) * (long long)nfeat + job.feature];
+      const bool left = (b == missing_bin) ? (job.default_left != 0) : (b <= job.split_bin);
+      ldest[i] = left ? atomicAdd(&lcnt, 1) : ~atomicAdd(&rcnt, 1);
+    }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      lbase = atomicAdd(&counters[j * 2], lcnt);
+      rbase = atomicAdd(&counters[j * 2 + 1], rcnt);
+    }
+    __syncthreads();
+    // copy records to their runs (left run ascending from start+lbase;
+    // right run descending from end-1-rbase)
+    for (int i = threadIdx.x; i < tile_n; i += blockDim.x) {
+      const int d = ldest[i];
+      const long long dst =
+          d >= 0 ? (long long)job.start + lbase + d : (long long)job.end - 1 - rbase - (~d);
+      const long long srow = tile + i;
+      dst_rows[dst] = src_rows[srow];
+      dst_gh[dst] = src_gh[srow];
+      const BinT* sp = src_bins + srow * (long long)nfeat;
+      BinT* dp = dst_bins + dst * (long long)nfeat;
+      if constexpr (sizeof(BinT) == 1) {
+        if ((nfeat & 3) == 0) {
+          const uchar4* sp4 = reinterpret_cast<const uchar4*>(sp);
+          uchar4* dp4 = reinterpret_cast<uchar4*>(dp);
+          for (int f4 = 0; f4 < (nfeat >> 2); ++f4) dp4[f4] = sp4[f4];
+          continue;
+        }
+      }
+      for (int f = 0; f < nfeat; ++f) dp[f] = sp[f];
+    }
+    __syncthreads();
+  }
+}
+
+// leaf scatter from compact row-id buffers
+__global__ __launch_bounds__(HIST_BLOCK) void leaf_update_compact_kernel(
+    const int* __restrict__ rows0, const int* __restrict__ rows1,
+    float* __restrict__ margin, const LeafJob* __restrict__ jobs,
+    const int* __restrict__ block_job, long long col_stride) {
+  const LeafJob job = jobs[block_job[blockIdx.x]];
+  const int* src = job.parity ? rows1 : rows0;
+  const int chunk = blockIdx.x - job.first_block;
+  const long long step = (long long)job.num_blocks * blockDim.x;
+  for (long long r = job.start + (long long)chunk * blockDim.x + threadIdx.x; r < job.end; r += step) {
+    margin[(long long)src[r] * col_stride] += job.value;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // split-gain scan
 //
 // Kernel A: one 256-thread workgroup per (node, feature): inclusive scan of
@@ -519,6 +665,62 @@ void predict_forest(torch::Tensor X, torch::Tensor left, torch::Tensor right,
                      (int)t_begin, (int)t_end, out.data_ptr<float>(), (int)k);
 }
 
+void hist_build_compact(torch::Tensor bins_c, torch::Tensor gh_c, torch::Tensor jobs,
+                        torch::Tensor block_job, torch::Tensor out, int64_t nfeat,
+                        int64_t stride, double scale_g, double scale_h, int64_t lds_words) {
+  CHECK_GPU(bins_c);
+  const int grid = (int)block_job.size(0);
+  const size_t lds_bytes = (size_t)lds_words * sizeof(unsigned long long);
+  auto stream = current_stream();
+  if (bins_c.scalar_type() == torch::kUInt8) {
+    hipLaunchKernelGGL(hist_compact_kernel<unsigned char>, dim3(grid), dim3(HIST_BLOCK), lds_bytes,
+                       stream, bins_c.data_ptr<unsigned char>(),
+                       (const float2*)gh_c.data_ptr<float>(), (const HistJob*)jobs.data_ptr<int>(),
+                       block_job.data_ptr<int>(), (unsigned long long*)out.data_ptr<int64_t>(),
+                       (int)nfeat, (int)stride, (float)scale_g, (float)scale_h);
+  } else {
+    hipLaunchKernelGGL(hist_compact_kernel<short>, dim3(grid), dim3(HIST_BLOCK), lds_bytes, stream,
+                       bins_c.data_ptr<short>(), (const float2*)gh_c.data_ptr<float>(),
+                       (const HistJob*)jobs.data_ptr<int>(), block_job.data_ptr<int>(),
+                       (unsigned long long*)out.data_ptr<int64_t>(), (int)nfeat, (int)stride,
+                       (float)scale_g, (float)scale_h);
+  }
+}
+
+void partition_compact(torch::Tensor src_bins, torch::Tensor src_gh, torch::Tensor src_rows,
+                       torch::Tensor dst_bins, torch::Tensor dst_gh, torch::Tensor dst_rows,
+                       torch::Tensor jobs, torch::Tensor block_job, torch::Tensor counters,
+                       int64_t nfeat, int64_t missing_bin) {
+  CHECK_GPU(src_bins);
+  const int grid = (int)block_job.size(0);
+  auto stream = current_stream();
+  if (src_bins.scalar_type() == torch::kUInt8) {
+    hipLaunchKernelGGL(partition_compact_kernel<unsigned char>, dim3(grid), dim3(HIST_BLOCK), 0,
+                       stream, src_bins.data_ptr<unsigned char>(),
+                       (const float2*)src_gh.data_ptr<float>(), src_rows.data_ptr<int>(),
+                       dst_bins.data_ptr<unsigned char>(), (float2*)dst_gh.data_ptr<float>(),
+                       dst_rows.data_ptr<int>(), (const PartJob*)jobs.data_ptr<int>(),
+                       block_job.data_ptr<int>(), counters.data_ptr<int>(), (int)nfeat,
+                       (int)missing_bin);
+  } else {
+    hipLaunchKernelGGL(partition_compact_kernel<short>, dim3(grid), dim3(HIST_BLOCK), 0, stream,
+                       src_bins.data_ptr<short>(), (const float2*)src_gh.data_ptr<float>(),
+                       src_rows.data_ptr<int>(), dst_bins.data_ptr<short>(),
+                       (float2*)dst_gh.data_ptr<float>(), dst_rows.data_ptr<int>(),
+                       (const PartJob*)jobs.data_ptr<int>(), block_job.data_ptr<int>(),
+                       counters.data_ptr<int>(), (int)nfeat, (int)missing_bin);
+  }
+}
+
+void leaf_update_compact(torch::Tensor rows0, torch::Tensor rows1, torch::Tensor margin_base,
+                         torch::Tensor jobs, torch::Tensor block_job, int64_t col_stride) {
+  CHECK_GPU(margin_base);
+  const int grid = (int)block_job.size(0);
+  hipLaunchKernelGGL(leaf_update_compact_kernel, dim3(grid), dim3(HIST_BLOCK), 0, current_stream(),
+                     rows0.data_ptr<int>(), rows1.data_ptr<int>(), margin_base.data_ptr<float>(),
+                     (const LeafJob*)jobs.data_ptr<int>(), block_job.data_ptr<int>(), col_stride);
+}
+
 void find_splits(torch::Tensor hist, torch::Tensor parent, torch::Tensor nbins,
                  torch::Tensor feat_mask, torch::Tensor monotone, torch::Tensor cands,
                  torch::Tensor out, int64_t k, int64_t f, int64_t stride, int64_t has_missing,
@@ -549,6 +751,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("hist_convert", &hist_convert, "fixed-point -> float32 histogram convert");
   m.def("partition", &partition, "batched two-ended row partition");
   m.def("find_splits", &find_splits, "fused split-gain scan + per-node reduce");
+  m.def("hist_build_compact", &hist_build_compact, "streaming histogram over compact row buffers");
+  m.def("partition_compact", &partition_compact, "compacting partition (rows+bins+gh rewrite)");
+  m.def("leaf_update_compact", &leaf_update_compact, "leaf scatter from compact row ids");
   m.def("leaf_update", &leaf_update, "batched leaf value scatter into margins");
   m.def("predict_forest", &predict_forest, "batched dense forest traversal");
   m.attr("_built_for") = "gfx950";

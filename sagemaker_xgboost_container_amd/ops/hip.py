@@ -270,3 +270,126 @@ def predict_forest_flat(flat, X, k, t_begin=0, t_end=None):
         t_begin, t_end, out, k,
     )
     return out
+
+
+class TreeState:
+    """Per-tree compact-layout state (v2 pipeline).
+
+    Rows live in node-contiguous compact SoA buffers (bins bytes, float2
+    gradient pairs, original row ids) that the compacting partition kernel
+    ping-pongs between; histogram builds stream them with perfectly
+    coalesced loads — no kernel gathers by row id after the root level.
+    Level 0 streams the original bin matrix directly (no setup copy) unless
+    the tree is row-subsampled.
+    """
+
+    def __init__(self, qm, gh, sample_rows=None):
+        self.qm = qm
+        device = qm.bins.device
+        n = qm.num_row
+        if sample_rows is None:
+            cap = n
+            self._rows_init = torch.arange(n, dtype=torch.int32, device=device)
+            self._bins_init = qm.bins
+            self._gh_init = gh.contiguous()
+        else:
+            cap = sample_rows.numel()
+            self._rows_init = sample_rows.to(torch.int32)
+            idx = sample_rows.long()
+            self._bins_init = qm.bins.index_select(0, idx).contiguous()
+            self._gh_init = gh.index_select(0, idx).contiguous()
+        self.cap = cap
+        cache = getattr(qm, "_compact_cache", None)
+        if cache is None or cache[0].shape[0] < cap:
+            cache = (
+                torch.empty((cap, qm.num_col), dtype=qm.bins.dtype, device=device),
+                torch.empty((cap, qm.num_col), dtype=qm.bins.dtype, device=device),
+                torch.empty((cap, 2), dtype=torch.float32, device=device),
+                torch.empty((cap, 2), dtype=torch.float32, device=device),
+                torch.empty(cap, dtype=torch.int32, device=device),
+                torch.empty(cap, dtype=torch.int32, device=device),
+            )
+            qm._compact_cache = cache
+        self._bins = [cache[0], cache[1]]
+        self._gh = [cache[2], cache[3]]
+        self._rows = [cache[4], cache[5]]
+        self._level0 = True  # parity-0 source is still the original matrix
+
+    def _src(self, parity):
+        if parity == 0 and self._level0:
+            return self._bins_init, self._gh_init, self._rows_init
+        return self._bins[parity], self._gh[parity], self._rows[parity]
+
+    def build_histograms(self, jobs, parity, scale):
+        qm = self.qm
+        f = qm.num_col
+        stride = qm.stride
+        bins_c, gh_c, _ = self._src(parity)
+        groups = _feature_groups(f, stride)
+        acc = torch.zeros((len(jobs), f * stride, 2), dtype=torch.int64, device=bins_c.device)
+        packed = []
+        blocks_per = []
+        first_block = 0
+        for hist_idx, (start, end) in enumerate(jobs):
+            rows = end - start
+            nb = int(max(1, min((rows + _ROWS_PER_BLOCK - 1) // _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB)))
+            for fg_start, fg_end in groups:
+                packed.append((start, end, hist_idx, fg_start, fg_end, first_block, nb))
+                blocks_per.append(nb)
+                first_block += nb
+        jobs_dev = _pack_jobs(packed)
+        block_job = _block_map(blocks_per)
+        lds_words = max((fe - fs) for fs, fe in groups) * stride * 2
+        _K.hist_build_compact(
+            bins_c, gh_c, jobs_dev, block_job, acc, f, stride, scale[0], scale[1], lds_words
+        )
+        return acc
+
+    def partition_level(self, segs, feats, split_bins, default_lefts, src_parity):
+        qm = self.qm
+        src_bins, src_gh, src_rows = self._src(src_parity)
+        dst = 1 - src_parity
+        packed = []
+        blocks_per = []
+        first_block = 0
+        for (start, end), feature, sbin, dl in zip(segs, feats, split_bins, default_lefts):
+            rows = end - start
+            nb = int(max(1, min((rows + _ROWS_PER_BLOCK - 1) // _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB)))
+            packed.append((start, end, int(feature), int(sbin), int(bool(dl)), first_block, nb))
+            blocks_per.append(nb)
+            first_block += nb
+        jobs_dev = _pack_jobs(packed)
+        block_job = _block_map(blocks_per)
+        counters = torch.zeros((len(segs), 2), dtype=torch.int32, device=src_bins.device)
+        missing_bin = qm.stride - 1 if qm.has_missing else -1
+        _K.partition_compact(
+            src_bins, src_gh, src_rows, self._bins[dst], self._gh[dst], self._rows[dst],
+            jobs_dev, block_job, counters, qm.num_col, missing_bin,
+        )
+        if src_parity == 0:
+            self._level0 = False
+        return counters[:, 0].cpu().tolist()
+
+    def update_margins(self, margin_col, leaf_jobs):
+        if not leaf_jobs:
+            return
+        rows0 = self._rows_init if self._level0 else self._rows[0]
+        packed = []
+        blocks_per = []
+        first_block = 0
+        for parity, start, end, value in leaf_jobs:
+            rows = end - start
+            nb = int(max(1, min((rows + _ROWS_PER_BLOCK - 1) // _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB)))
+            vbits = int(np.float32(value).view(np.int32))
+            packed.append((start, end, int(parity), vbits, first_block, nb))
+            blocks_per.append(nb)
+            first_block += nb
+        jobs_dev = _pack_jobs(packed)
+        block_job = _block_map(blocks_per)
+        _K.leaf_update_compact(
+            rows0, self._rows[1], margin_col, jobs_dev, block_job, margin_col.stride(0)
+        )
+
+
+def make_tree_state(qm, gh, sample_rows=None):
+    return TreeState(qm, gh, sample_rows)

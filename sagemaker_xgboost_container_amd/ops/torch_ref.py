@@ -284,3 +284,36 @@ def predict_forest_flat(flat, X, k, t_begin=0, t_end=None):
         active = left[node] >= 0
     out.index_add_(1, flat["tree_cls"][t_begin:t_end], flat["value"][node])
     return out
+
+
+class TreeState:
+    """Reference tree state: row-index ping-pong buffers (same contract as
+    the HIP compact-layout state)."""
+
+    def __init__(self, qm, gh, sample_rows=None):
+        self.qm = qm
+        self.gh = gh
+        device = qm.bins.device
+        rows = (
+            sample_rows.to(torch.int32)
+            if sample_rows is not None
+            else torch.arange(qm.num_row, dtype=torch.int32, device=device)
+        )
+        self.cap = rows.numel()
+        self._bufs = (rows.clone(), torch.empty_like(rows))
+
+    def build_histograms(self, jobs, parity, scale):
+        return build_histograms(self.qm, self.gh, self._bufs[parity], jobs, scale)
+
+    def partition_level(self, segs, feats, split_bins, default_lefts, src_parity):
+        return partition_level(
+            self.qm, self._bufs[src_parity], self._bufs[1 - src_parity],
+            segs, feats, split_bins, default_lefts,
+        )
+
+    def update_margins(self, margin_col, leaf_jobs):
+        update_margins(margin_col, self._bufs, leaf_jobs)
+
+
+def make_tree_state(qm, gh, sample_rows=None):
+    return TreeState(qm, gh, sample_rows)
