@@ -278,7 +278,18 @@ class HistGrower:
                 feature_mask=node_feature_mask,
                 monotone=self.monotone,
             )
-            packed = splits["packed"].cpu().numpy()  # one D2H transfer
+            # 3. partition every positive-gain segment against the DEVICE
+            # split tensor (no host round-trip before partitioning), then
+            # read splits + counts back in a single queue drain
+            parity = frontier[0].parity
+            counters = self.state.partition_level(
+                [node.seg for node in frontier],
+                list(range(len(frontier))),
+                splits["packed"],
+                parity,
+            )
+            packed = splits["packed"].cpu().numpy()  # the level's drain
+            counts = counters[:, 0].cpu().tolist()
             gains = packed[:, 0]
             feats = packed[:, 1].astype(int)
             bins = packed[:, 2].astype(int)
@@ -286,42 +297,28 @@ class HistGrower:
             lgs = packed[:, 4]
             lhs = packed[:, 5]
 
-            # 3. decide splits (leaf cap) and batch-partition
-            to_split = []
+            next_frontier = []
             for i, node in enumerate(frontier):
                 if gains[i] <= 0.0 or (p.max_leaves and n_leaves >= p.max_leaves):
                     finished.append(node)
-                else:
-                    to_split.append((i, node))
-                    n_leaves += 1
-
-            next_frontier = []
-            if to_split:
-                parity = to_split[0][1].parity
-                counts = self.state.partition_level(
-                    [node.seg for _i, node in to_split],
-                    [int(feats[i]) for i, _n in to_split],
-                    [int(bins[i]) for i, _n in to_split],
-                    [bool(dls[i]) for i, _n in to_split],
-                    parity,
+                    continue
+                n_leaves += 1
+                lid, rid, lstate, rstate = self._apply_split(
+                    tree, node, int(feats[i]), int(bins[i]), bool(dls[i]),
+                    float(gains[i]), float(lgs[i]), float(lhs[i]),
                 )
-                for (i, node), left_count in zip(to_split, counts):
-                    lid, rid, lstate, rstate = self._apply_split(
-                        tree, node, int(feats[i]), int(bins[i]), bool(dls[i]),
-                        float(gains[i]), float(lgs[i]), float(lhs[i]),
+                mid = node.start + counts[i]
+                next_frontier.append(
+                    _Node(lid, 1 - parity, node.start, mid, float(lgs[i]), float(lhs[i]),
+                          depth + 1, lstate[0], lstate[1], lstate[2])
+                )
+                next_frontier.append(
+                    _Node(
+                        rid, 1 - parity, mid, node.end,
+                        node.g - float(lgs[i]), node.h - float(lhs[i]), depth + 1,
+                        rstate[0], rstate[1], rstate[2],
                     )
-                    mid = node.start + left_count
-                    next_frontier.append(
-                        _Node(lid, 1 - parity, node.start, mid, float(lgs[i]), float(lhs[i]),
-                              depth + 1, lstate[0], lstate[1], lstate[2])
-                    )
-                    next_frontier.append(
-                        _Node(
-                            rid, 1 - parity, mid, node.end,
-                            node.g - float(lgs[i]), node.h - float(lhs[i]), depth + 1,
-                            rstate[0], rstate[1], rstate[2],
-                        )
-                    )
+                )
 
             prev_hists = hists
             prev_row = frontier_row
@@ -376,6 +373,7 @@ class HistGrower:
                             "left_g": float(row[4]),
                             "left_h": float(row[5]),
                             "gain": gain,
+                            "packed_dev": s["packed"],
                         },
                     ),
                 )
@@ -389,9 +387,10 @@ class HistGrower:
             node = nodes[nid]
             if p.max_depth and node.depth >= p.max_depth:
                 continue
-            counts = self.state.partition_level(
-                [node.seg], [s["feature"]], [s["bin"]], [s["default_left"]], node.parity
+            counters = self.state.partition_level(
+                [node.seg], [0], s["packed_dev"], node.parity
             )
+            counts = [int(counters[0, 0])]
             lid, rid, lstate, rstate = self._apply_split(
                 tree, node, s["feature"], s["bin"], s["default_left"], s["gain"], s["left_g"], s["left_h"]
             )

@@ -207,18 +207,29 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_compact_kernel(
 // destination run. Destinations inside one tile form two coalesced runs.
 #define CPART_TILE 2048
 
+// jobs carry segments only; the split decision (gain, feature, bin,
+// missing direction) is read from the on-device packed split tensor
+// ([k, 6] float32 from the split kernels), so no host round-trip sits
+// between split search and partition. gain <= 0 jobs are skipped.
 template <typename BinT>
 __global__ __launch_bounds__(HIST_BLOCK) void partition_compact_kernel(
     const BinT* __restrict__ src_bins, const float2* __restrict__ src_gh,
     const int* __restrict__ src_rows, BinT* __restrict__ dst_bins,
     float2* __restrict__ dst_gh, int* __restrict__ dst_rows,
     const PartJob* __restrict__ jobs, const int* __restrict__ block_job,
+    const float* __restrict__ split_packed,  // [n_nodes, 6]; job.feature = node row
     int* __restrict__ counters, int nfeat, int missing_bin) {
   __shared__ int ldest[CPART_TILE];  // destination index per tile row
   __shared__ int lcnt, rcnt, lbase, rbase;
 
   const int j = block_job[blockIdx.x];
   const PartJob job = jobs[j];
+  const float* sp6 = split_packed + job.feature * 6;
+  if (sp6[0] <= 0.0f) return;  // no split for this node
+  const int feature = (int)sp6[1];
+  const int split_bin = (int)sp6[2];
+  const int default_left = sp6[3] > 0.5f ? 1 : 0;
+
   const int chunk = blockIdx.x - job.first_block;
   const long long tile_step = (long long)job.num_blocks * CPART_TILE;
 
@@ -230,8 +241,8 @@ __global__ __launch_bounds__(HIST_BLOCK) void partition_compact_kernel(
     __syncthreads();
     const int tile_n = (int)min((long long)CPART_TILE, job.end - tile);
     for (int i = threadIdx.x; i < tile_n; i += blockDim.x) {
-      const int b = (int)src_bins[(tile + i) * (long long)nfeat + job.feature];
-      const bool left = (b == missing_bin) ? (job.default_left != 0) : (b <= job.split_bin);
+      const int b = (int)src_bins[(tile + i) * (long long)nfeat + feature];
+      const bool left = (b == missing_bin) ? (default_left != 0) : (b <= split_bin);
       ldest[i] = left ? atomicAdd(&lcnt, 1) : ~atomicAdd(&rcnt, 1);
     }
     __syncthreads();
@@ -240,26 +251,44 @@ __global__ __launch_bounds__(HIST_BLOCK) void partition_compact_kernel(
       rbase = atomicAdd(&counters[j * 2 + 1], rcnt);
     }
     __syncthreads();
-    // copy records to their runs (left run ascending from start+lbase;
-    // right run descending from end-1-rbase)
+    // copy phase, (row, dword) work units: consecutive threads move
+    // consecutive dwords so stores coalesce into the two destination runs
+    if ((nfeat & 3) == 0 && sizeof(BinT) == 1) {
+      const int nd = nfeat >> 2;  // bin dwords per row
+      const uchar4* sb4 = reinterpret_cast<const uchar4*>(src_bins);
+      uchar4* db4 = reinterpret_cast<uchar4*>(dst_bins);
+      for (int u = threadIdx.x; u < tile_n * nd; u += blockDim.x) {
+        const int i = u / nd;
+        const int f4 = u - i * nd;
+        const int d = ldest[i];
+        const long long dst =
+            d >= 0 ? (long long)job.start + lbase + d : (long long)job.end - 1 - rbase - (~d);
+        db4[dst * nd + f4] = sb4[(tile + i) * (long long)nd + f4];
+      }
+    } else {
+      for (int u = threadIdx.x; u < tile_n * nfeat; u += blockDim.x) {
+        const int i = u / nfeat;
+        const int f = u - i * nfeat;
+        const int d = ldest[i];
+        const long long dst =
+            d >= 0 ? (long long)job.start + lbase + d : (long long)job.end - 1 - rbase - (~d);
+        dst_bins[dst * (long long)nfeat + f] = src_bins[(tile + i) * (long long)nfeat + f];
+      }
+    }
+    for (int u = threadIdx.x; u < tile_n * 2; u += blockDim.x) {
+      const int i = u >> 1;
+      const int half = u & 1;
+      const int d = ldest[i];
+      const long long dst =
+          d >= 0 ? (long long)job.start + lbase + d : (long long)job.end - 1 - rbase - (~d);
+      reinterpret_cast<float*>(dst_gh)[dst * 2 + half] =
+          reinterpret_cast<const float*>(src_gh)[(tile + i) * 2 + half];
+    }
     for (int i = threadIdx.x; i < tile_n; i += blockDim.x) {
       const int d = ldest[i];
       const long long dst =
           d >= 0 ? (long long)job.start + lbase + d : (long long)job.end - 1 - rbase - (~d);
-      const long long srow = tile + i;
-      dst_rows[dst] = src_rows[srow];
-      dst_gh[dst] = src_gh[srow];
-      const BinT* sp = src_bins + srow * (long long)nfeat;
-      BinT* dp = dst_bins + dst * (long long)nfeat;
-      if constexpr (sizeof(BinT) == 1) {
-        if ((nfeat & 3) == 0) {
-          const uchar4* sp4 = reinterpret_cast<const uchar4*>(sp);
-          uchar4* dp4 = reinterpret_cast<uchar4*>(dp);
-          for (int f4 = 0; f4 < (nfeat >> 2); ++f4) dp4[f4] = sp4[f4];
-          continue;
-        }
-      }
-      for (int f = 0; f < nfeat; ++f) dp[f] = sp[f];
+      dst_rows[dst] = src_rows[tile + i];
     }
     __syncthreads();
   }
@@ -689,8 +718,8 @@ void hist_build_compact(torch::Tensor bins_c, torch::Tensor gh_c, torch::Tensor 
 
 void partition_compact(torch::Tensor src_bins, torch::Tensor src_gh, torch::Tensor src_rows,
                        torch::Tensor dst_bins, torch::Tensor dst_gh, torch::Tensor dst_rows,
-                       torch::Tensor jobs, torch::Tensor block_job, torch::Tensor counters,
-                       int64_t nfeat, int64_t missing_bin) {
+                       torch::Tensor jobs, torch::Tensor block_job, torch::Tensor split_packed,
+                       torch::Tensor counters, int64_t nfeat, int64_t missing_bin) {
   CHECK_GPU(src_bins);
   const int grid = (int)block_job.size(0);
   auto stream = current_stream();
@@ -700,15 +729,16 @@ void partition_compact(torch::Tensor src_bins, torch::Tensor src_gh, torch::Tens
                        (const float2*)src_gh.data_ptr<float>(), src_rows.data_ptr<int>(),
                        dst_bins.data_ptr<unsigned char>(), (float2*)dst_gh.data_ptr<float>(),
                        dst_rows.data_ptr<int>(), (const PartJob*)jobs.data_ptr<int>(),
-                       block_job.data_ptr<int>(), counters.data_ptr<int>(), (int)nfeat,
-                       (int)missing_bin);
+                       block_job.data_ptr<int>(), split_packed.data_ptr<float>(),
+                       counters.data_ptr<int>(), (int)nfeat, (int)missing_bin);
   } else {
     hipLaunchKernelGGL(partition_compact_kernel<short>, dim3(grid), dim3(HIST_BLOCK), 0, stream,
                        src_bins.data_ptr<short>(), (const float2*)src_gh.data_ptr<float>(),
                        src_rows.data_ptr<int>(), dst_bins.data_ptr<short>(),
                        (float2*)dst_gh.data_ptr<float>(), dst_rows.data_ptr<int>(),
                        (const PartJob*)jobs.data_ptr<int>(), block_job.data_ptr<int>(),
-                       counters.data_ptr<int>(), (int)nfeat, (int)missing_bin);
+                       split_packed.data_ptr<float>(), counters.data_ptr<int>(), (int)nfeat,
+                       (int)missing_bin);
   }
 }
 

@@ -345,17 +345,23 @@ class TreeState:
         )
         return acc
 
-    def partition_level(self, segs, feats, split_bins, default_lefts, src_parity):
+    def partition_level(self, segs, node_rows, split_packed, src_parity):
+        """Partition every segment whose on-device split has gain > 0.
+
+        node_rows[i] = row of segs[i] in split_packed ([k, 6] float32 from
+        find_splits, still on device — no host round-trip). Returns the
+        DEVICE counters tensor [J, 2]; the caller reads it back together
+        with the packed splits in one drain."""
         qm = self.qm
         src_bins, src_gh, src_rows = self._src(src_parity)
         dst = 1 - src_parity
         packed = []
         blocks_per = []
         first_block = 0
-        for (start, end), feature, sbin, dl in zip(segs, feats, split_bins, default_lefts):
+        for (start, end), node_row in zip(segs, node_rows):
             rows = end - start
             nb = int(max(1, min((rows + _ROWS_PER_BLOCK - 1) // _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB)))
-            packed.append((start, end, int(feature), int(sbin), int(bool(dl)), first_block, nb))
+            packed.append((start, end, int(node_row), 0, 0, first_block, nb))
             blocks_per.append(nb)
             first_block += nb
         jobs_dev = _pack_jobs(packed)
@@ -364,11 +370,11 @@ class TreeState:
         missing_bin = qm.stride - 1 if qm.has_missing else -1
         _K.partition_compact(
             src_bins, src_gh, src_rows, self._bins[dst], self._gh[dst], self._rows[dst],
-            jobs_dev, block_job, counters, qm.num_col, missing_bin,
+            jobs_dev, block_job, split_packed.contiguous(), counters, qm.num_col, missing_bin,
         )
         if src_parity == 0:
             self._level0 = False
-        return counters[:, 0].cpu().tolist()
+        return counters
 
     def update_margins(self, margin_col, leaf_jobs):
         if not leaf_jobs:

@@ -305,11 +305,28 @@ class TreeState:
     def build_histograms(self, jobs, parity, scale):
         return build_histograms(self.qm, self.gh, self._bufs[parity], jobs, scale)
 
-    def partition_level(self, segs, feats, split_bins, default_lefts, src_parity):
-        return partition_level(
+    def partition_level(self, segs, node_rows, split_packed, src_parity):
+        """Same contract as the HIP state: consume the packed split tensor,
+        skip gain <= 0 jobs, return a [J, 2] counters tensor."""
+        sp = split_packed.detach().cpu().numpy()
+        counters = torch.zeros((len(segs), 2), dtype=torch.int32)
+        do_segs, do_feats, do_bins, do_dls, rows_of = [], [], [], [], []
+        for j, ((start, end), node_row) in enumerate(zip(segs, node_rows)):
+            if sp[node_row, 0] <= 0.0:
+                continue
+            do_segs.append((start, end))
+            do_feats.append(int(sp[node_row, 1]))
+            do_bins.append(int(sp[node_row, 2]))
+            do_dls.append(bool(sp[node_row, 3] > 0.5))
+            rows_of.append(j)
+        counts = partition_level(
             self.qm, self._bufs[src_parity], self._bufs[1 - src_parity],
-            segs, feats, split_bins, default_lefts,
+            do_segs, do_feats, do_bins, do_dls,
         )
+        for j, c, (start, end) in zip(rows_of, counts, do_segs):
+            counters[j, 0] = c
+            counters[j, 1] = (end - start) - c
+        return counters
 
     def update_margins(self, margin_col, leaf_jobs):
         update_margins(margin_col, self._bufs, leaf_jobs)
