@@ -126,6 +126,20 @@ __global__ void hist_convert_kernel(const unsigned long long* __restrict__ in,
                                     float* __restrict__ out, long long n_pairs,
                                     float inv_g, float inv_h) {
   const long long step = (long long)gridDim.x * blockDim.x;
+  const long long i0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  for (long long i = i0; i < n_pairs; i += step) {
+    const long long j = i * 2;
+    out[j] = (float)((double)(long long)in[j] * (double)inv_g);
+    out[j + 1] = (float)((double)(long long)in[j + 1] * (double)inv_h);
+  }
+}
+
+__global__ void hist_convert_dev_kernel(const unsigned long long* __restrict__ in,
+                                        float* __restrict__ out, long long n_pairs,
+                                        const float* __restrict__ gh_max) {
+  const double inv_g = (double)fmaxf(gh_max[0], 1e-30f) / 8589934592.0;
+  const double inv_h = (double)fmaxf(gh_max[1], 1e-30f) / 8589934592.0;
+  const long long step = (long long)gridDim.x * blockDim.x;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n_pairs; i += step) {
     const long long j = i * 2;
     out[j] = (float)((double)(long long)in[j] * (double)inv_g);
@@ -150,9 +164,12 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_compact_kernel(
     const BinT* __restrict__ bins_c, const float2* __restrict__ gh_c,
     const HistJob* __restrict__ jobs, const int* __restrict__ block_job,
     unsigned long long* __restrict__ out, int nfeat, int stride,
-    float scale_g, float scale_h) {
+    const float* __restrict__ gh_max) {
   extern __shared__ unsigned long long lhist[];
 
+  // fixed-point scale derived on device (2^33 / max_abs): no host sync
+  const float scale_g = 8589934592.0f / fmaxf(gh_max[0], 1e-30f);
+  const float scale_h = 8589934592.0f / fmaxf(gh_max[1], 1e-30f);
   const HistJob job = jobs[block_job[blockIdx.x]];
   const int nf_group = job.fg_end - job.fg_start;
   const int lds_words = nf_group * stride * 2;
@@ -696,7 +713,7 @@ void predict_forest(torch::Tensor X, torch::Tensor left, torch::Tensor right,
 
 void hist_build_compact(torch::Tensor bins_c, torch::Tensor gh_c, torch::Tensor jobs,
                         torch::Tensor block_job, torch::Tensor out, int64_t nfeat,
-                        int64_t stride, double scale_g, double scale_h, int64_t lds_words) {
+                        int64_t stride, torch::Tensor gh_max, int64_t lds_words) {
   CHECK_GPU(bins_c);
   const int grid = (int)block_job.size(0);
   const size_t lds_bytes = (size_t)lds_words * sizeof(unsigned long long);
@@ -706,14 +723,23 @@ void hist_build_compact(torch::Tensor bins_c, torch::Tensor gh_c, torch::Tensor 
                        stream, bins_c.data_ptr<unsigned char>(),
                        (const float2*)gh_c.data_ptr<float>(), (const HistJob*)jobs.data_ptr<int>(),
                        block_job.data_ptr<int>(), (unsigned long long*)out.data_ptr<int64_t>(),
-                       (int)nfeat, (int)stride, (float)scale_g, (float)scale_h);
+                       (int)nfeat, (int)stride, gh_max.data_ptr<float>());
   } else {
     hipLaunchKernelGGL(hist_compact_kernel<short>, dim3(grid), dim3(HIST_BLOCK), lds_bytes, stream,
                        bins_c.data_ptr<short>(), (const float2*)gh_c.data_ptr<float>(),
                        (const HistJob*)jobs.data_ptr<int>(), block_job.data_ptr<int>(),
                        (unsigned long long*)out.data_ptr<int64_t>(), (int)nfeat, (int)stride,
-                       (float)scale_g, (float)scale_h);
+                       gh_max.data_ptr<float>());
   }
+}
+
+void hist_convert_dev(torch::Tensor in, torch::Tensor out, torch::Tensor gh_max) {
+  CHECK_GPU(in);
+  const long long n_pairs = in.numel() / 2;
+  const int grid = (int)std::min<long long>((n_pairs + HIST_BLOCK - 1) / HIST_BLOCK, 2048);
+  hipLaunchKernelGGL(hist_convert_dev_kernel, dim3(std::max(grid, 1)), dim3(HIST_BLOCK), 0,
+                     current_stream(), (const unsigned long long*)in.data_ptr<int64_t>(),
+                     out.data_ptr<float>(), n_pairs, gh_max.data_ptr<float>());
 }
 
 void partition_compact(torch::Tensor src_bins, torch::Tensor src_gh, torch::Tensor src_rows,
@@ -782,6 +808,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("partition", &partition, "batched two-ended row partition");
   m.def("find_splits", &find_splits, "fused split-gain scan + per-node reduce");
   m.def("hist_build_compact", &hist_build_compact, "streaming histogram over compact row buffers");
+  m.def("hist_convert_dev", &hist_convert_dev, "fixed-point -> f32 convert with device-resident scale");
   m.def("partition_compact", &partition_compact, "compacting partition (rows+bins+gh rewrite)");
   m.def("leaf_update_compact", &leaf_update_compact, "leaf scatter from compact row ids");
   m.def("leaf_update", &leaf_update, "batched leaf value scatter into margins");

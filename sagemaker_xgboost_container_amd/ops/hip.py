@@ -38,13 +38,12 @@ _MAX_BLOCKS_PER_JOB = int(_os.environ.get("SMXGB_MAX_BLOCKS", "512"))
 
 
 def compute_scale(gh, comm=None):
-    m = gh.abs().amax(dim=0)  # (2,)
+    """Device-resident (gmax, hmax); kernels derive 2^33/max on device —
+    no host synchronization per tree."""
+    m = gh.abs().amax(dim=0).contiguous()  # (2,)
     if comm is not None:
         comm.allreduce_max_(m)
-    gmax, hmax = (float(v) for v in m.cpu())
-    scale_g = 2.0**_FIXED_BITS / max(gmax, 1e-30)
-    scale_h = 2.0**_FIXED_BITS / max(hmax, 1e-30)
-    return (scale_g, scale_h)
+    return m
 
 
 def _feature_groups(nfeat, stride):
@@ -95,6 +94,9 @@ def build_histograms(qm, gh, rowbuf, jobs, scale):
     jobs_dev = _pack_jobs(packed)
     block_job = _block_map(blocks_per)
     lds_words = max((fe - fs) for fs, fe in groups) * stride * 2
+    if isinstance(scale, torch.Tensor):  # device-resident (gmax, hmax)
+        gmax, hmax = (float(v) for v in scale.cpu())
+        scale = (2.0**_FIXED_BITS / max(gmax, 1e-30), 2.0**_FIXED_BITS / max(hmax, 1e-30))
     _K.hist_build(
         qm.bins, gh.contiguous(), rowbuf, jobs_dev, block_job, acc,
         f, stride, scale[0], scale[1], lds_words,
@@ -104,7 +106,10 @@ def build_histograms(qm, gh, rowbuf, jobs, scale):
 
 def hist_to_float(acc, scale):
     out = torch.empty(acc.shape, dtype=torch.float32, device=acc.device)
-    _K.hist_convert(acc, out, 1.0 / scale[0], 1.0 / scale[1])
+    if isinstance(scale, torch.Tensor):
+        _K.hist_convert_dev(acc, out, scale)
+    else:
+        _K.hist_convert(acc, out, 1.0 / scale[0], 1.0 / scale[1])
     return out
 
 
@@ -341,7 +346,7 @@ class TreeState:
         block_job = _block_map(blocks_per)
         lds_words = max((fe - fs) for fs, fe in groups) * stride * 2
         _K.hist_build_compact(
-            bins_c, gh_c, jobs_dev, block_job, acc, f, stride, scale[0], scale[1], lds_words
+            bins_c, gh_c, jobs_dev, block_job, acc, f, stride, scale, lds_words
         )
         return acc
 
