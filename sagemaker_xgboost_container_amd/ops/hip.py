@@ -402,5 +402,52 @@ class TreeState:
         )
 
 
+class _V1TreeState:
+    """Gather-based (v1) pipeline kept for measured A/B comparisons
+    (SMXGB_PIPELINE=v1)."""
+
+    def __init__(self, qm, gh, sample_rows=None):
+        self.qm = qm
+        self.gh = gh.contiguous()
+        device = qm.bins.device
+        rows = (
+            sample_rows.to(torch.int32)
+            if sample_rows is not None
+            else torch.arange(qm.num_row, dtype=torch.int32, device=device)
+        )
+        self.cap = rows.numel()
+        self._bufs = (rows.clone(), torch.empty_like(rows))
+
+    def build_histograms(self, jobs, parity, scale):
+        return build_histograms(self.qm, self.gh, self._bufs[parity], jobs, scale)
+
+    def partition_level(self, segs, node_rows, split_packed, src_parity):
+        sp = split_packed.cpu().numpy()
+        counters = torch.zeros((len(segs), 2), dtype=torch.int32)
+        do_segs, do_feats, do_bins, do_dls, rows_of = [], [], [], [], []
+        for j, ((start, end), node_row) in enumerate(zip(segs, node_rows)):
+            if sp[node_row, 0] <= 0.0:
+                continue
+            do_segs.append((start, end))
+            do_feats.append(int(sp[node_row, 1]))
+            do_bins.append(int(sp[node_row, 2]))
+            do_dls.append(bool(sp[node_row, 3] > 0.5))
+            rows_of.append(j)
+        if do_segs:
+            counts = partition_level(
+                self.qm, self._bufs[src_parity], self._bufs[1 - src_parity],
+                do_segs, do_feats, do_bins, do_dls,
+            )
+            for j, c, (start, end) in zip(rows_of, counts, do_segs):
+                counters[j, 0] = c
+                counters[j, 1] = (end - start) - c
+        return counters
+
+    def update_margins(self, margin_col, leaf_jobs):
+        update_margins(margin_col, self._bufs, leaf_jobs)
+
+
 def make_tree_state(qm, gh, sample_rows=None):
+    if _os.environ.get("SMXGB_PIPELINE") == "v1":
+        return _V1TreeState(qm, gh, sample_rows)
     return TreeState(qm, gh, sample_rows)
