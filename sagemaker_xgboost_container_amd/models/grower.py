@@ -197,6 +197,7 @@ class HistGrower:
             leaf_jobs = self._grow_lossguide(tree, gh, scale, root_node, tree_mask)
         else:
             leaf_jobs = self._grow_depthwise(tree, gh, scale, root_node, tree_mask)
+        tree.finalize()
         return tree, leaf_jobs
 
 

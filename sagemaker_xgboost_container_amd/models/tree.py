@@ -1,25 +1,57 @@
-"""Regression-tree structure (structure-of-arrays, xgboost-layout-compatible).
+"""Regression-tree structure (xgboost-layout-compatible).
 
 Node arrays use xgboost conventions so serialization maps 1:1 onto the
 Booster JSON schema ("left_children"/"right_children"/"split_indices"/
 "split_conditions"/"default_left"/"base_weights"/"loss_changes"/
 "sum_hessian"): leaf iff left < 0; test `fvalue < threshold` goes left.
+
+Construction uses Python lists (O(1) append — the grower adds two nodes per
+split in the hot loop); `finalize()` converts to numpy arrays once when the
+tree is done.
 """
 import numpy as np
 
 
 class Tree:
     def __init__(self):
-        self.left = np.zeros(0, dtype=np.int32)
-        self.right = np.zeros(0, dtype=np.int32)
-        self.parent = np.zeros(0, dtype=np.int32)
-        self.feature = np.zeros(0, dtype=np.int32)
-        self.threshold = np.zeros(0, dtype=np.float32)   # split condition (fvalue < t -> left)
-        self.split_bin = np.zeros(0, dtype=np.int32)     # training-time bin index of the split
-        self.default_left = np.zeros(0, dtype=bool)
-        self.value = np.zeros(0, dtype=np.float32)       # leaf value / base weight
-        self.gain = np.zeros(0, dtype=np.float32)
-        self.sum_hess = np.zeros(0, dtype=np.float32)
+        self.left = []
+        self.right = []
+        self.parent = []
+        self.feature = []
+        self.threshold = []      # split condition (fvalue < t -> left)
+        self.split_bin = []      # training-time bin index of the split
+        self.default_left = []
+        self.value = []          # leaf value / base weight
+        self.gain = []
+        self.sum_hess = []
+        self._finalized = False
+
+    _FIELDS = (
+        ("left", np.int32),
+        ("right", np.int32),
+        ("parent", np.int32),
+        ("feature", np.int32),
+        ("threshold", np.float32),
+        ("split_bin", np.int32),
+        ("default_left", bool),
+        ("value", np.float32),
+        ("gain", np.float32),
+        ("sum_hess", np.float32),
+    )
+
+    def finalize(self):
+        """Convert construction lists to numpy arrays (idempotent)."""
+        if not self._finalized:
+            for name, dtype in self._FIELDS:
+                setattr(self, name, np.asarray(getattr(self, name), dtype=dtype))
+            self._finalized = True
+        return self
+
+    def _editable(self):
+        if self._finalized:
+            for name, _dtype in self._FIELDS:
+                setattr(self, name, list(getattr(self, name)))
+            self._finalized = False
 
     @property
     def num_nodes(self):
@@ -30,21 +62,24 @@ class Tree:
 
     @property
     def num_leaves(self):
-        return int((self.left < 0).sum())
+        if self._finalized:
+            return int((self.left < 0).sum())
+        return sum(1 for v in self.left if v < 0)
 
     def add_node(self, parent=-1, value=0.0, sum_hess=0.0):
         """Append a leaf node; returns its id."""
-        nid = self.num_nodes
-        self.left = np.append(self.left, np.int32(-1))
-        self.right = np.append(self.right, np.int32(-1))
-        self.parent = np.append(self.parent, np.int32(parent))
-        self.feature = np.append(self.feature, np.int32(0))
-        self.threshold = np.append(self.threshold, np.float32(0.0))
-        self.split_bin = np.append(self.split_bin, np.int32(-1))
-        self.default_left = np.append(self.default_left, False)
-        self.value = np.append(self.value, np.float32(value))
-        self.gain = np.append(self.gain, np.float32(0.0))
-        self.sum_hess = np.append(self.sum_hess, np.float32(sum_hess))
+        self._editable()
+        nid = len(self.left)
+        self.left.append(-1)
+        self.right.append(-1)
+        self.parent.append(parent)
+        self.feature.append(0)
+        self.threshold.append(0.0)
+        self.split_bin.append(-1)
+        self.default_left.append(False)
+        self.value.append(float(value))
+        self.gain.append(0.0)
+        self.sum_hess.append(float(sum_hess))
         return nid
 
     def apply_split(self, nid, feature, threshold, split_bin, default_left, gain,
@@ -54,11 +89,11 @@ class Tree:
         rid = self.add_node(parent=nid, value=right_value, sum_hess=right_hess)
         self.left[nid] = lid
         self.right[nid] = rid
-        self.feature[nid] = feature
-        self.threshold[nid] = threshold
-        self.split_bin[nid] = split_bin
-        self.default_left[nid] = default_left
-        self.gain[nid] = gain
+        self.feature[nid] = int(feature)
+        self.threshold[nid] = float(threshold)
+        self.split_bin[nid] = int(split_bin)
+        self.default_left[nid] = bool(default_left)
+        self.gain[nid] = float(gain)
         return lid, rid
 
     def depth(self, nid):
@@ -84,6 +119,7 @@ class Tree:
 
     def to_arrays(self):
         """Flat dict of node arrays (device-upload / serialization form)."""
+        self.finalize()
         return {
             "left": self.left,
             "right": self.right,
@@ -110,4 +146,5 @@ class Tree:
         t.value = np.asarray(arrays["value"], dtype=np.float32)
         t.gain = np.asarray(arrays.get("gain", np.zeros(n)), dtype=np.float32)
         t.sum_hess = np.asarray(arrays.get("sum_hess", np.zeros(n)), dtype=np.float32)
+        t._finalized = True
         return t
