@@ -269,23 +269,33 @@ __global__ __launch_bounds__(HIST_BLOCK) void partition_compact_kernel(
     }
     __syncthreads();
     // copy phase, (row, dword) work units: consecutive threads move
-    // consecutive dwords so stores coalesce into the two destination runs
+    // consecutive dwords so stores coalesce into the two destination runs.
+    // Work units are padded to a power of two per row so the row index is a
+    // shift, not an integer division (a div per dword dominated this loop).
     if ((nfeat & 3) == 0 && sizeof(BinT) == 1) {
       const int nd = nfeat >> 2;  // bin dwords per row
+      int log2p = 0;
+      while ((1 << log2p) < nd) ++log2p;
+      const int mask = (1 << log2p) - 1;
       const uchar4* sb4 = reinterpret_cast<const uchar4*>(src_bins);
       uchar4* db4 = reinterpret_cast<uchar4*>(dst_bins);
-      for (int u = threadIdx.x; u < tile_n * nd; u += blockDim.x) {
-        const int i = u / nd;
-        const int f4 = u - i * nd;
+      for (int u = threadIdx.x; u < (tile_n << log2p); u += blockDim.x) {
+        const int i = u >> log2p;
+        const int f4 = u & mask;
+        if (f4 >= nd) continue;
         const int d = ldest[i];
         const long long dst =
             d >= 0 ? (long long)job.start + lbase + d : (long long)job.end - 1 - rbase - (~d);
         db4[dst * nd + f4] = sb4[(tile + i) * (long long)nd + f4];
       }
     } else {
-      for (int u = threadIdx.x; u < tile_n * nfeat; u += blockDim.x) {
-        const int i = u / nfeat;
-        const int f = u - i * nfeat;
+      int log2p = 0;
+      while ((1 << log2p) < nfeat) ++log2p;
+      const int mask = (1 << log2p) - 1;
+      for (int u = threadIdx.x; u < (tile_n << log2p); u += blockDim.x) {
+        const int i = u >> log2p;
+        const int f = u & mask;
+        if (f >= nfeat) continue;
         const int d = ldest[i];
         const long long dst =
             d >= 0 ? (long long)job.start + lbase + d : (long long)job.end - 1 - rbase - (~d);

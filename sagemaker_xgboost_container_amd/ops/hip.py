@@ -72,6 +72,21 @@ def _block_map(blocks_per_job):
     return torch.from_numpy(bj).cuda(non_blocking=True)
 
 
+def _pack_jobs_and_map(fields_list, blocks_per_job):
+    """One fused H2D upload: [job table | block->job map]."""
+    width = len(fields_list[0])
+    total_blocks = int(sum(blocks_per_job))
+    arr = np.empty(len(fields_list) * width + total_blocks, dtype=np.int32)
+    jobs_flat = np.asarray(fields_list, dtype=np.int32).reshape(-1)
+    arr[: jobs_flat.size] = jobs_flat
+    ofs = jobs_flat.size
+    for j, nb in enumerate(blocks_per_job):
+        arr[ofs : ofs + nb] = j
+        ofs += nb
+    dev = torch.from_numpy(arr).cuda(non_blocking=True)
+    return dev[: jobs_flat.size], dev[jobs_flat.size :]
+
+
 def build_histograms(qm, gh, rowbuf, jobs, scale):
     f = qm.num_col
     stride = qm.stride
@@ -342,8 +357,7 @@ class TreeState:
                 packed.append((start, end, hist_idx, fg_start, fg_end, first_block, nb))
                 blocks_per.append(nb)
                 first_block += nb
-        jobs_dev = _pack_jobs(packed)
-        block_job = _block_map(blocks_per)
+        jobs_dev, block_job = _pack_jobs_and_map(packed, blocks_per)
         lds_words = max((fe - fs) for fs, fe in groups) * stride * 2
         _K.hist_build_compact(
             bins_c, gh_c, jobs_dev, block_job, acc, f, stride, scale, lds_words
@@ -369,8 +383,7 @@ class TreeState:
             packed.append((start, end, int(node_row), 0, 0, first_block, nb))
             blocks_per.append(nb)
             first_block += nb
-        jobs_dev = _pack_jobs(packed)
-        block_job = _block_map(blocks_per)
+        jobs_dev, block_job = _pack_jobs_and_map(packed, blocks_per)
         counters = torch.zeros((len(segs), 2), dtype=torch.int32, device=src_bins.device)
         missing_bin = qm.stride - 1 if qm.has_missing else -1
         _K.partition_compact(
