@@ -148,7 +148,12 @@ def run_module(module_dir, cmd_args, env_vars, entry_point, capture_error=False)
 
     env = dict(os.environ)
     env.update({k: str(v) for k, v in (env_vars or {}).items()})
-    env["PYTHONPATH"] = code_dir + os.pathsep + env.get("PYTHONPATH", "")
+    # child sees the user code dir plus this framework (when running
+    # in-tree rather than pip-installed)
+    framework_root = os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    env["PYTHONPATH"] = os.pathsep.join(
+        [code_dir, framework_root, env.get("PYTHONPATH", "")]
+    )
 
     if script.endswith(".py"):
         cmd = [sys.executable, script] + list(cmd_args or [])

@@ -1,0 +1,101 @@
+"""Script-mode training: user entry point executed with SageMaker env/args
+(the reference's test_abalone.py script-mode scenarios, in-process)."""
+import json
+import os
+
+import numpy as np
+import pytest
+
+from sagemaker_xgboost_container_amd import training
+from sagemaker_xgboost_container_amd.constants import sm_env_constants as smc
+from sagemaker_xgboost_container_amd.models.booster import Booster
+from sagemaker_xgboost_container_amd.utils import sm_env
+
+
+@pytest.fixture
+def script_env(tmp_opt_ml, monkeypatch, tmp_path):
+    base = tmp_opt_ml
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(300, 6))
+    y = X[:, 0] * 2 + rng.normal(scale=0.1, size=300)
+    lines = [
+        f"{y[i]:.5f} " + " ".join(f"{j}:{X[i, j]:.5f}" for j in range(6)) for i in range(300)
+    ]
+    (base / "input/data/train/data.libsvm").write_text("\n".join(lines))
+
+    repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code_dir = tmp_path / "code"
+    code_dir.mkdir()
+    src = open(os.path.join(repo_root, "examples", "abalone_script_mode.py")).read()
+    (code_dir / "train_script.py").write_text(src)
+
+    hp = {
+        "sagemaker_program": "train_script.py",
+        "sagemaker_submit_directory": str(code_dir),
+        "num_round": "5",
+        "max_depth": "3",
+    }
+    (base / "input/config/hyperparameters.json").write_text(json.dumps(hp))
+    (base / "input/config/inputdataconfig.json").write_text(
+        json.dumps({"train": {"ContentType": "libsvm", "TrainingInputMode": "File",
+                              "S3DistributionType": "FullyReplicated"}})
+    )
+    (base / "input/config/resourceconfig.json").write_text(
+        json.dumps({"hosts": ["algo-1"], "current_host": "algo-1"})
+    )
+
+    monkeypatch.setenv(smc.SM_INPUT_TRAINING_CONFIG_FILE, str(base / "input/config/hyperparameters.json"))
+    monkeypatch.setenv(smc.SM_INPUT_DATA_CONFIG_FILE, str(base / "input/config/inputdataconfig.json"))
+    monkeypatch.setenv(smc.SM_CHECKPOINT_CONFIG_FILE, str(base / "input/config/checkpointconfig.json"))
+    monkeypatch.setenv(smc.SM_CHANNEL_TRAIN, str(base / "input/data/train"))
+    monkeypatch.setenv(smc.SM_HOSTS, '["algo-1"]')
+    monkeypatch.setenv(smc.SM_CURRENT_HOST, "algo-1")
+    monkeypatch.setenv(smc.SM_MODEL_DIR, str(base / "model"))
+    monkeypatch.setenv(smc.SM_OUTPUT_DATA_DIR, str(base / "output/data"))
+    monkeypatch.delenv("SAGEMAKER_PROGRAM", raising=False)
+    return base
+
+
+def test_script_mode_trains_and_saves(script_env):
+    env = sm_env.TrainingEnv(base_path=str(script_env))
+    assert env.user_entry_point == "train_script.py"
+    training.train(env)
+    model_path = script_env / "model" / "xgboost-model"
+    assert model_path.exists()
+    bst = Booster()
+    bst.load_model(model_path)
+    assert bst.num_boosted_rounds() == 5
+
+
+def test_script_mode_env_vars_passed(script_env, tmp_path):
+    # a script that dumps its env proves the SM_* contract reaches the child
+    code_dir = tmp_path / "code2"
+    code_dir.mkdir()
+    out_file = tmp_path / "env.json"
+    (code_dir / "probe.py").write_text(
+        "import json, os, sys\n"
+        f"json.dump({{'hps': os.environ.get('SM_HPS'), 'train': os.environ.get('SM_CHANNEL_TRAIN'),"
+        f" 'args': sys.argv[1:]}}, open({str(out_file)!r}, 'w'))\n"
+    )
+    hp_file = script_env / "input/config/hyperparameters.json"
+    hp = json.loads(hp_file.read_text())
+    hp["sagemaker_program"] = "probe.py"
+    hp["sagemaker_submit_directory"] = str(code_dir)
+    hp_file.write_text(json.dumps(hp))
+
+    env = sm_env.TrainingEnv(base_path=str(script_env))
+    training.train(env)
+    probe = json.loads(out_file.read_text())
+    assert "num_round" in probe["hps"]
+    assert "--num_round" in probe["args"]
+    assert probe["train"].endswith("input/data/train")
+
+
+def test_algorithm_mode_when_no_program(script_env):
+    hp_file = script_env / "input/config/hyperparameters.json"
+    hp = {"num_round": "3", "objective": "reg:squarederror"}
+    hp_file.write_text(json.dumps(hp))
+    env = sm_env.TrainingEnv(base_path=str(script_env))
+    assert env.user_entry_point is None
+    training.train(env)
+    assert (script_env / "model" / "xgboost-model").exists()
