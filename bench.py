@@ -78,7 +78,8 @@ def main():
     world = comm.world_size if comm else 1
     rank = comm.rank if comm else 0
     if use_gpu:
-        device = torch.device("cuda", int(os.environ.get("LOCAL_RANK", 0)))
+        local = int(os.environ.get("LOCAL_RANK", 0)) % max(torch.cuda.device_count(), 1)
+        device = torch.device("cuda", local)
         torch.cuda.set_device(device)
     else:
         device = torch.device("cpu")

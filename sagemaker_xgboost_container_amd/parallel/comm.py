@@ -64,9 +64,11 @@ def init_from_env(backend=None, timeout_s=1800):
     if world_size <= 1:
         return None
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = os.environ.get("SMXGB_COMM_BACKEND") or (
+            "nccl" if torch.cuda.is_available() else "gloo"
+        )
     local_rank = int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
-    if backend == "nccl":
+    if torch.cuda.is_available():
         torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
     dist.init_process_group(backend=backend, timeout=datetime.timedelta(seconds=timeout_s))
     return Communicator()
