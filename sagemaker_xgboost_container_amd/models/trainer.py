@@ -177,6 +177,22 @@ def train(
         cbs.append(EarlyStopping(rounds=early_stopping_rounds, maximize=bool(maximize)))
     container = CallbackContainer(cbs)
 
+    # process_type=update: run refresh/prune updaters over the EXISTING
+    # trees instead of growing new ones (xgboost semantics; reference HP
+    # surface algorithm_mode/hyperparameter_validation.py updater_validator)
+    if params.get("process_type") == "update":
+        from .refresh import run_update_process
+
+        booster = container.before_training(booster)
+        run_update_process(
+            booster, params, X, y, weight, objective, margin, n_outputs,
+            num_boost_round, container, eval_sets, metric_names, feval, comm,
+        )
+        booster = container.after_training(booster)
+        if evals_result is not None:
+            evals_result.update(container.history)
+        return booster
+
     booster_kind = params.get("booster", "gbtree")
     if booster_kind == "gblinear":
         from .gblinear import LinearModel, LinearUpdater

@@ -256,9 +256,32 @@ def make_flat_forest(trees, tree_info, weight_drop, device):
     }
 
 
+def _native_predict(flat, X, k, t_begin, t_end):
+    """Parallel C++ traversal (serving hot path on CPU hosts)."""
+    from . import _smxgb_hip as K
+
+    if "_i32" not in flat:
+        flat["_i32"] = {
+            "left": flat["left"].to(torch.int32).contiguous(),
+            "right": flat["right"].to(torch.int32).contiguous(),
+            "feature": flat["feature"].to(torch.int32).contiguous(),
+            "default_left": flat["default_left"].to(torch.uint8).contiguous(),
+            "tree_root": flat["tree_root"].to(torch.int32).contiguous(),
+            "tree_cls": flat["tree_cls"].to(torch.int32).contiguous(),
+        }
+    f = flat["_i32"]
+    out = torch.zeros((X.shape[0], k), dtype=torch.float32)
+    K.predict_forest_cpu(
+        X.contiguous(), f["left"], f["right"], f["feature"],
+        flat["threshold"].contiguous(), f["default_left"], flat["value"].contiguous(),
+        f["tree_root"], f["tree_cls"], t_begin, t_end, out, k,
+    )
+    return out
+
+
 def predict_forest_flat(flat, X, k, t_begin=0, t_end=None):
-    """(n, k) margin contributions of trees [t_begin, t_end) — vectorized
-    level-synchronized traversal over all (row, tree) pairs."""
+    """(n, k) margin contributions of trees [t_begin, t_end) — native C++
+    traversal when the extension is built, else vectorized torch."""
     if t_end is None:
         t_end = flat["n_trees"]
     n = X.shape[0]
@@ -266,6 +289,11 @@ def predict_forest_flat(flat, X, k, t_begin=0, t_end=None):
     out = torch.zeros((n, k), dtype=torch.float32, device=X.device)
     if T <= 0:
         return out
+    if X.device.type == "cpu":
+        try:
+            return _native_predict(flat, X, k, t_begin, t_end)
+        except ImportError:
+            pass
     node = flat["tree_root"][t_begin:t_end].unsqueeze(0).expand(n, T).contiguous()
     left = flat["left"]
     right = flat["right"]
