@@ -179,6 +179,26 @@ class HistGrower:
         cap = self.state.cap
 
         scale = self.backend.compute_scale(gh, comm=self.comm)
+        tree_mask = self._sample_features(p.colsample_bytree, None)
+
+        # device-autonomous fast path: the whole depthwise tree enqueued
+        # with zero host syncs (one readback at the end)
+        import os as _os
+
+        if (
+            hasattr(self.backend, "DeviceGrower")
+            and _os.environ.get("SMXGB_NO_DEVICE_GROW") != "1"
+            and p.grow_policy == "depthwise"
+            and p.max_leaves == 0
+            and 1 <= p.max_depth <= 10
+            and self.monotone is None
+            and self.inter_sets is None
+            and p.colsample_bylevel >= 1.0
+            and p.colsample_bynode >= 1.0
+        ):
+            tree, leaf_jobs = self._grow_depthwise_device(gh, scale, cap, tree_mask)
+            tree.finalize()
+            return tree, leaf_jobs
 
         tree = Tree()
         root_sum = (
@@ -191,13 +211,65 @@ class HistGrower:
         root = tree.add_node(parent=-1, value=self._weight(G, H) * p.eta, sum_hess=H)
         root_node = _Node(root, 0, 0, cap, G, H, 0)
 
-        tree_mask = self._sample_features(p.colsample_bytree, None)
-
         if p.grow_policy == "lossguide":
             leaf_jobs = self._grow_lossguide(tree, gh, scale, root_node, tree_mask)
         else:
             leaf_jobs = self._grow_depthwise(tree, gh, scale, root_node, tree_mask)
         tree.finalize()
+        return tree, leaf_jobs
+
+    def _grow_depthwise_device(self, gh, scale, cap, tree_mask):
+        """Consume the DeviceGrower's one-readback result into a Tree."""
+        p = self.p
+        qm = self.qm
+        D = p.max_depth
+        key = (D, id(self.state.qm))
+        if getattr(self, "_device_grower_key", None) != key:
+            self._device_grower = self.backend.DeviceGrower(self.state, D)
+            self._device_grower_key = key
+        dg = self._device_grower
+        dg.state = self.state  # fresh per-tree compact state
+        if tree_mask is not None:
+            dg.mask = tree_mask.to(torch.uint8).contiguous()
+        splits_np, counts_np, root_np = dg.grow(
+            scale, (p.reg_lambda, p.reg_alpha, p.gamma, p.min_child_weight), self.comm
+        )
+        self.state._level0 = not bool(splits_np[0, 0] > 0)
+
+        if not hasattr(qm, "_cuts_np"):
+            qm._cuts_np = qm.cuts.cpu().numpy()
+            qm._cut_ptr_np = qm.cut_ptr.cpu().numpy()
+
+        tree = Tree()
+        G, H = float(root_np[0]), float(root_np[1])
+        root = tree.add_node(parent=-1, value=self._weight(G, H) * p.eta, sum_hess=H)
+        leaf_jobs = []
+        stack = [(0, 0, root, 0, cap, G, H)]
+        while stack:
+            d, i, nid, start, end, g, h = stack.pop()
+            hidx = (1 << d) - 1 + i
+            gain = float(splits_np[hidx, 0]) if d < D else -1.0
+            if d < D and gain > 0.0:
+                feat = int(splits_np[hidx, 1])
+                sbin = int(splits_np[hidx, 2])
+                dl = bool(splits_np[hidx, 3] > 0.5)
+                lg = float(splits_np[hidx, 4])
+                lh = float(splits_np[hidx, 5])
+                lc = int(counts_np[hidx, 0])
+                threshold = float(qm._cuts_np[int(qm._cut_ptr_np[feat]) + sbin])
+                lid, rid = tree.apply_split(
+                    nid, feat, threshold, sbin, dl, gain,
+                    left_value=self._weight(lg, lh) * p.eta,
+                    right_value=self._weight(g - lg, h - lh) * p.eta,
+                    left_hess=lh,
+                    right_hess=h - lh,
+                )
+                mid = start + lc
+                stack.append((d + 1, 2 * i, lid, start, mid, lg, lh))
+                stack.append((d + 1, 2 * i + 1, rid, mid, end, g - lg, h - lh))
+            else:
+                parity = 0 if d == 0 else d % 2
+                leaf_jobs.append((parity, start, end, float(tree.value[nid])))
         return tree, leaf_jobs
 
 

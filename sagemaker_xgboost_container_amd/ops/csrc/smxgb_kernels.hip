@@ -336,6 +336,320 @@ __global__ __launch_bounds__(HIST_BLOCK) void leaf_update_compact_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// DEVICE-AUTONOMOUS depthwise grow (v3)
+//
+// The per-level host round-trip (drain -> python job packing -> launches)
+// costs ~2-3 ms of a 7 ms round. v3 removes it: the whole tree's kernel
+// sequence is enqueued up front; a tiny single-block `make_level` kernel
+// builds the next level's node table and block-assignment prefixes ON
+// DEVICE, hist/partition kernels walk virtual-block work lists found by
+// binary search, and the host reads back ONE packed buffer per tree
+// (split records + child counts + node sums) to build the tree object.
+//
+// Heap indexing: level d slot i lives at heap index (2^d - 1 + i);
+// children of (d, i) are (d+1, 2i) and (d+1, 2i+1). Dead slots have
+// start == end.
+// ---------------------------------------------------------------------------
+
+struct LevelNode {
+  int start;
+  int end;
+  int build;  // 1 = histogram built from rows, 0 = derived by subtraction
+};
+
+// prefix tables for one level's virtual-block work distribution
+struct LevelWork {
+  int hist_total;   // total hist virtual blocks
+  int part_total;   // total partition virtual blocks
+};
+
+#define GROW_MAX_SLOTS 1024  // max nodes per level for the device driver
+
+// Single-block kernel: build level d+1's node table from level d's
+// splits/counts, choose build children (smaller hessian), compute
+// per-slot virtual-block counts + exclusive prefixes for hist & partition.
+__global__ __launch_bounds__(GROW_MAX_SLOTS) void make_level_kernel(
+    const LevelNode* __restrict__ cur,    // [k] level d nodes
+    const float* __restrict__ splits,     // [k, 6] level d split records
+    const int* __restrict__ counts,       // [k, 2] level d partition counts
+    const float* __restrict__ node_gh,    // [k, 2] level d node sums
+    LevelNode* __restrict__ nxt,          // [2k] level d+1 nodes
+    float* __restrict__ nxt_gh,           // [2k, 2]
+    int* __restrict__ hist_prefix,        // [2k + 1]
+    int* __restrict__ part_prefix,        // [2k + 1]
+    LevelWork* __restrict__ work, int k, int rows_per_block, int max_blocks) {
+  __shared__ int h_counts[2 * GROW_MAX_SLOTS];
+  __shared__ int p_counts[2 * GROW_MAX_SLOTS];
+  const int tid = threadIdx.x;
+
+  for (int i = tid; i < k; i += blockDim.x) {
+    const float* sp = splits + i * 6;
+    const LevelNode node = cur[i];
+    LevelNode lft = {0, 0, 0}, rgt = {0, 0, 0};
+    float lg = 0.f, lh = 0.f, rg = 0.f, rh = 0.f;
+    if (node.end > node.start && sp[0] > 0.0f) {
+      const int lc = counts[i * 2];
+      lft.start = node.start;
+      lft.end = node.start + lc;
+      rgt.start = lft.end;
+      rgt.end = node.end;
+      lg = sp[4];
+      lh = sp[5];
+      rg = node_gh[i * 2] - lg;
+      rh = node_gh[i * 2 + 1] - lh;
+      // subtraction trick: build the smaller-hessian child
+      if (lh <= rh) {
+        lft.build = 1;
+      } else {
+        rgt.build = 1;
+      }
+    }
+    nxt[2 * i] = lft;
+    nxt[2 * i + 1] = rgt;
+    nxt_gh[4 * i] = lg;
+    nxt_gh[4 * i + 1] = lh;
+    nxt_gh[4 * i + 2] = rg;
+    nxt_gh[4 * i + 3] = rh;
+    for (int c = 0; c < 2; ++c) {
+      const LevelNode child = c ? rgt : lft;
+      const int rows = child.end - child.start;
+      h_counts[2 * i + c] =
+          (child.build && rows > 0)
+              ? (int)min((long long)(rows + rows_per_block - 1) / rows_per_block, (long long)max_blocks)
+              : 0;
+      p_counts[2 * i + c] =
+          rows > 0 ? (int)min((long long)(rows + rows_per_block - 1) / rows_per_block,
+                              (long long)max_blocks)
+                   : 0;
+    }
+  }
+  __syncthreads();
+  if (tid == 0) {  // serial scan: 2k <= 2048 entries, negligible
+    int hs = 0, ps = 0;
+    for (int i = 0; i < 2 * k; ++i) {
+      hist_prefix[i] = hs;
+      part_prefix[i] = ps;
+      hs += h_counts[i];
+      ps += p_counts[i];
+    }
+    hist_prefix[2 * k] = hs;
+    part_prefix[2 * k] = ps;
+    work->hist_total = hs;
+    work->part_total = ps;
+  }
+}
+
+// Root bootstrap: one node covering [0, cap); also its hist job prefix.
+__global__ void make_root_kernel(LevelNode* root, int* hist_prefix, int* part_prefix,
+                                 LevelWork* work, int cap, int rows_per_block, int max_blocks) {
+  root->start = 0;
+  root->end = cap;
+  root->build = 1;
+  const int nb = (int)min((long long)(cap + rows_per_block - 1) / rows_per_block, (long long)max_blocks);
+  hist_prefix[0] = 0;
+  hist_prefix[1] = nb;
+  part_prefix[0] = 0;
+  part_prefix[1] = nb;
+  work->hist_total = nb;
+  work->part_total = nb;
+}
+
+__device__ inline int find_slot(const int* __restrict__ prefix, int k, int vb) {
+  int lo = 0, hi = k;  // find i with prefix[i] <= vb < prefix[i+1]
+  while (lo + 1 < hi) {
+    const int mid = (lo + hi) >> 1;
+    if (prefix[mid] <= vb) {
+      lo = mid;
+    } else {
+      hi = mid;
+    }
+  }
+  return lo;
+}
+
+// hist over compact buffers driven by a device job table
+template <typename BinT>
+__global__ __launch_bounds__(HIST_BLOCK) void hist_device_kernel(
+    const BinT* __restrict__ bins_c, const float2* __restrict__ gh_c,
+    const LevelNode* __restrict__ nodes, const int* __restrict__ hist_prefix,
+    const LevelWork* __restrict__ work, unsigned long long* __restrict__ out,
+    int k, int nfeat, int stride, int n_groups, int feats_per_group,
+    const float* __restrict__ gh_max, int rows_per_block) {
+  extern __shared__ unsigned long long lhist[];
+  const float scale_g = 8589934592.0f / fmaxf(gh_max[0], 1e-30f);
+  const float scale_h = 8589934592.0f / fmaxf(gh_max[1], 1e-30f);
+  const int total = work->hist_total * n_groups;
+
+  for (int vb = blockIdx.x; vb < total; vb += gridDim.x) {
+    const int fg = vb % n_groups;
+    const int hvb = vb / n_groups;
+    const int slot = find_slot(hist_prefix, k, hvb);
+    const int chunk = hvb - hist_prefix[slot];
+    const int nb = hist_prefix[slot + 1] - hist_prefix[slot];
+    const LevelNode node = nodes[slot];
+    const int fg_start = fg * feats_per_group;
+    const int nf_group = min(feats_per_group, nfeat - fg_start);
+    const int lds_words = nf_group * stride * 2;
+
+    for (int i = threadIdx.x; i < lds_words; i += blockDim.x) lhist[i] = 0ull;
+    __syncthreads();
+
+    const long long step = (long long)nb * blockDim.x;
+    for (long long r = node.start + (long long)chunk * blockDim.x + threadIdx.x; r < node.end;
+         r += step) {
+      const float2 gp = gh_c[r];
+      const unsigned long long gfix = (unsigned long long)(long long)llrintf(gp.x * scale_g);
+      const unsigned long long hfix = (unsigned long long)(long long)llrintf(gp.y * scale_h);
+      const BinT* rp = bins_c + (long long)r * nfeat + fg_start;
+      #pragma unroll 4
+      for (int f = 0; f < nf_group; ++f) {
+        const int slot2 = (f * stride + (int)rp[f]) * 2;
+        atomicAdd(&lhist[slot2], gfix);
+        atomicAdd(&lhist[slot2 + 1], hfix);
+      }
+    }
+    __syncthreads();
+    unsigned long long* gout =
+        out + ((long long)slot * nfeat + fg_start) * (long long)stride * 2;
+    for (int i = threadIdx.x; i < lds_words; i += blockDim.x) {
+      const unsigned long long v = lhist[i];
+      if (v) atomicAdd(&gout[i], v);
+    }
+    __syncthreads();
+  }
+}
+
+// convert built slots (int64 acc -> f32 heap) and fill node sums from the
+// feature-0 bin range; one block per (slot, chunk of slots_total)
+__global__ __launch_bounds__(HIST_BLOCK) void convert_level_kernel(
+    const unsigned long long* __restrict__ acc, float* __restrict__ hist_f32,
+    const LevelNode* __restrict__ nodes, int k, long long slots2,
+    const float* __restrict__ gh_max) {
+  const double inv_g = (double)fmaxf(gh_max[0], 1e-30f) / 8589934592.0;
+  const double inv_h = (double)fmaxf(gh_max[1], 1e-30f) / 8589934592.0;
+  for (long long u = (long long)blockIdx.x * blockDim.x + threadIdx.x; u < (long long)k * slots2;
+       u += (long long)gridDim.x * blockDim.x) {
+    const int slot = (int)(u / slots2);
+    if (!nodes[slot].build || nodes[slot].end <= nodes[slot].start) continue;
+    const long long j = u - (long long)slot * slots2;
+    const double inv = (j & 1) ? inv_h : inv_g;
+    hist_f32[u] = (float)((double)(long long)acc[u] * inv);
+  }
+}
+
+// derived slots: hist = parent - sibling (all f32, same heap layout);
+// also node sums for EVERY alive slot from the feature-0 bins
+__global__ __launch_bounds__(HIST_BLOCK) void derive_level_kernel(
+    float* __restrict__ level_hist, const float* __restrict__ parent_hist,
+    const LevelNode* __restrict__ nodes, int k, long long slots2) {
+  for (long long u = (long long)blockIdx.x * blockDim.x + threadIdx.x; u < (long long)k * slots2;
+       u += (long long)gridDim.x * blockDim.x) {
+    const int slot = (int)(u / slots2);
+    const LevelNode node = nodes[slot];
+    if (node.build || node.end <= node.start) continue;
+    const long long j = u - (long long)slot * slots2;
+    const int sib = slot ^ 1;
+    const int parent = slot >> 1;
+    level_hist[u] = parent_hist[(long long)parent * slots2 + j] -
+                    level_hist[(long long)sib * slots2 + j];
+  }
+}
+
+// partition driven by the device job table (virtual blocks)
+template <typename BinT>
+__global__ __launch_bounds__(HIST_BLOCK) void partition_device_kernel(
+    const BinT* __restrict__ src_bins, const float2* __restrict__ src_gh,
+    const int* __restrict__ src_rows, BinT* __restrict__ dst_bins,
+    float2* __restrict__ dst_gh, int* __restrict__ dst_rows,
+    const LevelNode* __restrict__ nodes, const int* __restrict__ part_prefix,
+    const LevelWork* __restrict__ work, const float* __restrict__ split_packed,
+    int* __restrict__ counters, int k, int nfeat, int missing_bin) {
+  __shared__ int ldest[CPART_TILE];
+  __shared__ int lcnt, rcnt, lbase, rbase;
+  const int total = work->part_total;
+
+  for (int vb = blockIdx.x; vb < total; vb += gridDim.x) {
+    const int slot = find_slot(part_prefix, k, vb);
+    const LevelNode node = nodes[slot];
+    const float* sp6 = split_packed + slot * 6;
+    if (sp6[0] <= 0.0f || node.end <= node.start) continue;
+    const int feature = (int)sp6[1];
+    const int split_bin = (int)sp6[2];
+    const int default_left = sp6[3] > 0.5f ? 1 : 0;
+    const int chunk = vb - part_prefix[slot];
+    const int nb = part_prefix[slot + 1] - part_prefix[slot];
+    const long long tile_step = (long long)nb * CPART_TILE;
+
+    for (long long tile = node.start + (long long)chunk * CPART_TILE; tile < node.end;
+         tile += tile_step) {
+      if (threadIdx.x == 0) {
+        lcnt = 0;
+        rcnt = 0;
+      }
+      __syncthreads();
+      const int tile_n = (int)min((long long)CPART_TILE, node.end - tile);
+      for (int i = threadIdx.x; i < tile_n; i += blockDim.x) {
+        const int b = (int)src_bins[(tile + i) * (long long)nfeat + feature];
+        const bool left = (b == missing_bin) ? (default_left != 0) : (b <= split_bin);
+        ldest[i] = left ? atomicAdd(&lcnt, 1) : ~atomicAdd(&rcnt, 1);
+      }
+      __syncthreads();
+      if (threadIdx.x == 0) {
+        lbase = atomicAdd(&counters[slot * 2], lcnt);
+        rbase = atomicAdd(&counters[slot * 2 + 1], rcnt);
+      }
+      __syncthreads();
+      if ((nfeat & 3) == 0 && sizeof(BinT) == 1) {
+        const int nd = nfeat >> 2;
+        int log2p = 0;
+        while ((1 << log2p) < nd) ++log2p;
+        const int mask = (1 << log2p) - 1;
+        const uchar4* sb4 = reinterpret_cast<const uchar4*>(src_bins);
+        uchar4* db4 = reinterpret_cast<uchar4*>(dst_bins);
+        for (int u = threadIdx.x; u < (tile_n << log2p); u += blockDim.x) {
+          const int i = u >> log2p;
+          const int f4 = u & mask;
+          if (f4 >= nd) continue;
+          const int d = ldest[i];
+          const long long dst =
+              d >= 0 ? (long long)node.start + lbase + d : (long long)node.end - 1 - rbase - (~d);
+          db4[dst * nd + f4] = sb4[(tile + i) * (long long)nd + f4];
+        }
+      } else {
+        int log2p = 0;
+        while ((1 << log2p) < nfeat) ++log2p;
+        const int mask = (1 << log2p) - 1;
+        for (int u = threadIdx.x; u < (tile_n << log2p); u += blockDim.x) {
+          const int i = u >> log2p;
+          const int f = u & mask;
+          if (f >= nfeat) continue;
+          const int d = ldest[i];
+          const long long dst =
+              d >= 0 ? (long long)node.start + lbase + d : (long long)node.end - 1 - rbase - (~d);
+          dst_bins[dst * (long long)nfeat + f] = src_bins[(tile + i) * (long long)nfeat + f];
+        }
+      }
+      for (int u = threadIdx.x; u < tile_n * 2; u += blockDim.x) {
+        const int i = u >> 1;
+        const int half = u & 1;
+        const int d = ldest[i];
+        const long long dst =
+            d >= 0 ? (long long)node.start + lbase + d : (long long)node.end - 1 - rbase - (~d);
+        reinterpret_cast<float*>(dst_gh)[dst * 2 + half] =
+            reinterpret_cast<const float*>(src_gh)[(tile + i) * 2 + half];
+      }
+      for (int i = threadIdx.x; i < tile_n; i += blockDim.x) {
+        const int d = ldest[i];
+        const long long dst =
+            d >= 0 ? (long long)node.start + lbase + d : (long long)node.end - 1 - rbase - (~d);
+        dst_rows[dst] = src_rows[tile + i];
+      }
+      __syncthreads();
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // split-gain scan
 //
 // Kernel A: one 256-thread workgroup per (node, feature): inclusive scan of
@@ -787,6 +1101,100 @@ void leaf_update_compact(torch::Tensor rows0, torch::Tensor rows1, torch::Tensor
                      (const LeafJob*)jobs.data_ptr<int>(), block_job.data_ptr<int>(), col_stride);
 }
 
+void grow_make_root(torch::Tensor nodes, torch::Tensor hist_prefix, torch::Tensor part_prefix,
+                    torch::Tensor work, int64_t cap, int64_t rows_per_block, int64_t max_blocks) {
+  hipLaunchKernelGGL(make_root_kernel, dim3(1), dim3(1), 0, current_stream(),
+                     (LevelNode*)nodes.data_ptr<int>(), hist_prefix.data_ptr<int>(),
+                     part_prefix.data_ptr<int>(), (LevelWork*)work.data_ptr<int>(), (int)cap,
+                     (int)rows_per_block, (int)max_blocks);
+}
+
+void grow_make_level(torch::Tensor cur, torch::Tensor splits, torch::Tensor counts,
+                     torch::Tensor node_gh, torch::Tensor nxt, torch::Tensor nxt_gh,
+                     torch::Tensor hist_prefix, torch::Tensor part_prefix, torch::Tensor work,
+                     int64_t k, int64_t rows_per_block, int64_t max_blocks) {
+  hipLaunchKernelGGL(make_level_kernel, dim3(1), dim3(GROW_MAX_SLOTS), 0, current_stream(),
+                     (const LevelNode*)cur.data_ptr<int>(), splits.data_ptr<float>(),
+                     counts.data_ptr<int>(), node_gh.data_ptr<float>(),
+                     (LevelNode*)nxt.data_ptr<int>(), nxt_gh.data_ptr<float>(),
+                     hist_prefix.data_ptr<int>(), part_prefix.data_ptr<int>(),
+                     (LevelWork*)work.data_ptr<int>(), (int)k, (int)rows_per_block,
+                     (int)max_blocks);
+}
+
+void grow_hist_level(torch::Tensor bins_c, torch::Tensor gh_c, torch::Tensor nodes,
+                     torch::Tensor hist_prefix, torch::Tensor work, torch::Tensor acc,
+                     int64_t k, int64_t nfeat, int64_t stride, int64_t n_groups,
+                     int64_t feats_per_group, torch::Tensor gh_max, int64_t rows_per_block,
+                     int64_t grid, int64_t lds_words) {
+  const size_t lds_bytes = (size_t)lds_words * sizeof(unsigned long long);
+  auto stream = current_stream();
+  if (bins_c.scalar_type() == torch::kUInt8) {
+    hipLaunchKernelGGL(hist_device_kernel<unsigned char>, dim3((int)grid), dim3(HIST_BLOCK),
+                       lds_bytes, stream, bins_c.data_ptr<unsigned char>(),
+                       (const float2*)gh_c.data_ptr<float>(),
+                       (const LevelNode*)nodes.data_ptr<int>(), hist_prefix.data_ptr<int>(),
+                       (const LevelWork*)work.data_ptr<int>(),
+                       (unsigned long long*)acc.data_ptr<int64_t>(), (int)k, (int)nfeat,
+                       (int)stride, (int)n_groups, (int)feats_per_group,
+                       gh_max.data_ptr<float>(), (int)rows_per_block);
+  } else {
+    hipLaunchKernelGGL(hist_device_kernel<short>, dim3((int)grid), dim3(HIST_BLOCK), lds_bytes,
+                       stream, bins_c.data_ptr<short>(), (const float2*)gh_c.data_ptr<float>(),
+                       (const LevelNode*)nodes.data_ptr<int>(), hist_prefix.data_ptr<int>(),
+                       (const LevelWork*)work.data_ptr<int>(),
+                       (unsigned long long*)acc.data_ptr<int64_t>(), (int)k, (int)nfeat,
+                       (int)stride, (int)n_groups, (int)feats_per_group,
+                       gh_max.data_ptr<float>(), (int)rows_per_block);
+  }
+}
+
+void grow_convert_level(torch::Tensor acc, torch::Tensor hist_f32, torch::Tensor nodes,
+                        int64_t k, int64_t slots2, torch::Tensor gh_max) {
+  const long long total = (long long)k * slots2;
+  const int grid = (int)std::min<long long>((total + HIST_BLOCK - 1) / HIST_BLOCK, 2048);
+  hipLaunchKernelGGL(convert_level_kernel, dim3(std::max(grid, 1)), dim3(HIST_BLOCK), 0,
+                     current_stream(), (const unsigned long long*)acc.data_ptr<int64_t>(),
+                     hist_f32.data_ptr<float>(), (const LevelNode*)nodes.data_ptr<int>(), (int)k,
+                     slots2, gh_max.data_ptr<float>());
+}
+
+void grow_derive_level(torch::Tensor level_hist, torch::Tensor parent_hist, torch::Tensor nodes,
+                       int64_t k, int64_t slots2) {
+  const long long total = (long long)k * slots2;
+  const int grid = (int)std::min<long long>((total + HIST_BLOCK - 1) / HIST_BLOCK, 2048);
+  hipLaunchKernelGGL(derive_level_kernel, dim3(std::max(grid, 1)), dim3(HIST_BLOCK), 0,
+                     current_stream(), level_hist.data_ptr<float>(),
+                     parent_hist.data_ptr<float>(), (const LevelNode*)nodes.data_ptr<int>(),
+                     (int)k, slots2);
+}
+
+void grow_partition_level(torch::Tensor src_bins, torch::Tensor src_gh, torch::Tensor src_rows,
+                          torch::Tensor dst_bins, torch::Tensor dst_gh, torch::Tensor dst_rows,
+                          torch::Tensor nodes, torch::Tensor part_prefix, torch::Tensor work,
+                          torch::Tensor split_packed, torch::Tensor counters, int64_t k,
+                          int64_t nfeat, int64_t missing_bin, int64_t grid) {
+  auto stream = current_stream();
+  if (src_bins.scalar_type() == torch::kUInt8) {
+    hipLaunchKernelGGL(partition_device_kernel<unsigned char>, dim3((int)grid), dim3(HIST_BLOCK),
+                       0, stream, src_bins.data_ptr<unsigned char>(),
+                       (const float2*)src_gh.data_ptr<float>(), src_rows.data_ptr<int>(),
+                       dst_bins.data_ptr<unsigned char>(), (float2*)dst_gh.data_ptr<float>(),
+                       dst_rows.data_ptr<int>(), (const LevelNode*)nodes.data_ptr<int>(),
+                       part_prefix.data_ptr<int>(), (const LevelWork*)work.data_ptr<int>(),
+                       split_packed.data_ptr<float>(), counters.data_ptr<int>(), (int)k,
+                       (int)nfeat, (int)missing_bin);
+  } else {
+    hipLaunchKernelGGL(partition_device_kernel<short>, dim3((int)grid), dim3(HIST_BLOCK), 0,
+                       stream, src_bins.data_ptr<short>(), (const float2*)src_gh.data_ptr<float>(),
+                       src_rows.data_ptr<int>(), dst_bins.data_ptr<short>(),
+                       (float2*)dst_gh.data_ptr<float>(), dst_rows.data_ptr<int>(),
+                       (const LevelNode*)nodes.data_ptr<int>(), part_prefix.data_ptr<int>(),
+                       (const LevelWork*)work.data_ptr<int>(), split_packed.data_ptr<float>(),
+                       counters.data_ptr<int>(), (int)k, (int)nfeat, (int)missing_bin);
+  }
+}
+
 void find_splits(torch::Tensor hist, torch::Tensor parent, torch::Tensor nbins,
                  torch::Tensor feat_mask, torch::Tensor monotone, torch::Tensor cands,
                  torch::Tensor out, int64_t k, int64_t f, int64_t stride, int64_t has_missing,
@@ -819,6 +1227,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("find_splits", &find_splits, "fused split-gain scan + per-node reduce");
   m.def("hist_build_compact", &hist_build_compact, "streaming histogram over compact row buffers");
   m.def("hist_convert_dev", &hist_convert_dev, "fixed-point -> f32 convert with device-resident scale");
+  m.def("grow_make_root", &grow_make_root);
+  m.def("grow_make_level", &grow_make_level);
+  m.def("grow_hist_level", &grow_hist_level);
+  m.def("grow_convert_level", &grow_convert_level);
+  m.def("grow_derive_level", &grow_derive_level);
+  m.def("grow_partition_level", &grow_partition_level);
   m.def("partition_compact", &partition_compact, "compacting partition (rows+bins+gh rewrite)");
   m.def("leaf_update_compact", &leaf_update_compact, "leaf scatter from compact row ids");
   m.def("leaf_update", &leaf_update, "batched leaf value scatter into margins");

@@ -460,6 +460,134 @@ class _V1TreeState:
         update_margins(margin_col, self._bufs, leaf_jobs)
 
 
+
+_GROW_HIST_GRID = int(_os.environ.get("SMXGB_GROW_HIST_GRID", "768"))
+_GROW_PART_GRID = int(_os.environ.get("SMXGB_GROW_PART_GRID", "1024"))
+
+
+class DeviceGrower:
+    """v3: the whole depthwise tree enqueued with ZERO host syncs; one
+    readback (splits + counts heaps) per tree.
+
+    Level tables, block-assignment prefixes and the sibling-subtraction
+    choice are computed on device by `make_level`; hist/partition kernels
+    walk virtual-block work lists via binary search. Python time between
+    enqueues overlaps GPU execution.
+    """
+
+    def __init__(self, state, max_depth, feature_mask=None):
+        assert 1 <= max_depth <= 10, "device grower supports max_depth 1..10"
+        self.state = state
+        self.qm = state.qm
+        self.D = max_depth
+        qm = state.qm
+        device = qm.bins.device
+        f = qm.num_col
+        stride = qm.stride
+        self.slots2 = f * stride * 2
+        H = (1 << max_depth) - 1           # nodes in levels 0..D-1
+        max_k = 1 << (max_depth - 1)
+        self.H = H
+        self.nodes = torch.empty((H, 3), dtype=torch.int32, device=device)
+        self.node_gh = torch.zeros((H, 2), dtype=torch.float32, device=device)
+        self.splits = torch.empty((H, 6), dtype=torch.float32, device=device)
+        self.counts = torch.zeros((H, 2), dtype=torch.int32, device=device)
+        self.hist_f32 = torch.empty((H, self.slots2), dtype=torch.float32, device=device)
+        self.acc = torch.empty((max_k, self.slots2), dtype=torch.int64, device=device)
+        self.cands = torch.empty((max_k, f, 5), dtype=torch.float32, device=device)
+        self.hp = [torch.empty((1 << d) + 1, dtype=torch.int32, device=device) for d in range(max_depth)]
+        self.pp = [torch.empty((1 << d) + 1, dtype=torch.int32, device=device) for d in range(max_depth)]
+        self.work = torch.empty((max_depth, 2), dtype=torch.int32, device=device)
+        groups = _feature_groups(f, stride)
+        self.n_groups = len(groups)
+        self.feats_per_group = groups[0][1] - groups[0][0]
+        self.lds_words = max(fe - fs for fs, fe in groups) * stride * 2
+        if feature_mask is None:
+            self.mask = torch.empty(0, dtype=torch.uint8, device=device)
+        else:
+            self.mask = feature_mask.to(torch.uint8).contiguous()
+        self.mono = torch.empty(0, dtype=torch.int8, device=device)
+        if not hasattr(qm, "_nbins_i32"):
+            qm._nbins_i32 = qm.nbins.to(torch.int32).contiguous()
+
+    def grow(self, scale, split_params, comm=None):
+        """Enqueue the full tree; returns (splits_np [H,6], counts_np [H,2],
+        root_gh_np [2])."""
+        st = self.state
+        qm = self.qm
+        f = qm.num_col
+        stride = qm.stride
+        missing_bin = stride - 1 if qm.has_missing else -1
+        reg_lambda, reg_alpha, gamma, mcw = split_params
+
+        self.counts.zero_()
+        root_gh = st._gh_init.to(torch.float64).sum(0)
+        if comm is not None:
+            comm.allreduce_(root_gh)
+        self.node_gh[0] = root_gh.to(torch.float32)
+
+        for d in range(self.D):
+            k = 1 << d
+            base = k - 1
+            nodes_d = self.nodes[base : base + k]
+            gh_d = self.node_gh[base : base + k]
+            splits_d = self.splits[base : base + k]
+            counts_d = self.counts[base : base + k]
+
+            if d == 0:
+                _K.grow_make_root(nodes_d, self.hp[0], self.pp[0], self.work[0],
+                                  st.cap, _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB)
+            else:
+                pk = k >> 1
+                pbase = pk - 1
+                _K.grow_make_level(
+                    self.nodes[pbase : pbase + pk], self.splits[pbase : pbase + pk],
+                    self.counts[pbase : pbase + pk], self.node_gh[pbase : pbase + pk],
+                    nodes_d, gh_d, self.hp[d], self.pp[d], self.work[d],
+                    pk, _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB,
+                )
+
+            src_bins, src_gh, src_rows = (
+                (st._bins_init, st._gh_init, st._rows_init) if d == 0
+                else (st._bins[d % 2], st._gh[d % 2], st._rows[d % 2])
+            )
+            acc_d = self.acc[:k]
+            acc_d.zero_()
+            _K.grow_hist_level(
+                src_bins, src_gh, nodes_d, self.hp[d], self.work[d], acc_d,
+                k, f, stride, self.n_groups, self.feats_per_group, scale,
+                _ROWS_PER_BLOCK, _GROW_HIST_GRID, self.lds_words,
+            )
+            if comm is not None:
+                comm.allreduce_(acc_d)
+            hist_d = self.hist_f32[base : base + k]
+            _K.grow_convert_level(acc_d, hist_d, nodes_d, k, self.slots2, scale)
+            if d > 0:
+                pbase = (k >> 1) - 1
+                _K.grow_derive_level(hist_d, self.hist_f32[pbase : pbase + (k >> 1)],
+                                     nodes_d, k, self.slots2)
+
+            _K.find_splits(
+                hist_d, gh_d, qm._nbins_i32, self.mask, self.mono,
+                self.cands[:k], splits_d, k, f, stride,
+                1 if qm.has_missing else 0, 0, reg_lambda, reg_alpha, gamma, mcw,
+            )
+
+            dst = 1 - d % 2
+            _K.grow_partition_level(
+                src_bins, src_gh, src_rows, st._bins[dst], st._gh[dst], st._rows[dst],
+                nodes_d, self.pp[d], self.work[d], splits_d, counts_d,
+                k, f, missing_bin, _GROW_PART_GRID,
+            )
+        # (the grower decides _level0 from whether the root split)
+
+        # the tree's single host drain
+        splits_np = self.splits.cpu().numpy()
+        counts_np = self.counts.cpu().numpy()
+        root_np = self.node_gh[0].cpu().numpy()
+        return splits_np, counts_np, root_np
+
+
 def make_tree_state(qm, gh, sample_rows=None):
     if _os.environ.get("SMXGB_PIPELINE") == "v1":
         return _V1TreeState(qm, gh, sample_rows)

@@ -247,3 +247,60 @@ class TestEndToEnd:
             verbose_eval=False,
         )
         assert res["train"]["mlogloss"][-1] < res["train"]["mlogloss"][0]
+
+
+class TestDeviceGrower:
+    def test_matches_per_level_path(self, monkeypatch):
+        """The zero-sync device-autonomous grower must produce the same
+        ensemble as the per-level path (same kernels, same order)."""
+        import json as _json
+
+        from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+        from sagemaker_xgboost_container_amd.models import trainer
+
+        rng = np.random.default_rng(11)
+        X = rng.normal(size=(300_000, 28)).astype(np.float32)
+        y = (X[:, 0] * 2 - X[:, 1] + 0.5 * X[:, 2] * X[:, 3] > 0).astype(np.float32)
+        params = {"objective": "binary:logistic", "max_depth": 6, "eta": 0.3, "device": "cuda"}
+
+        monkeypatch.setenv("SMXGB_NO_DEVICE_GROW", "1")
+        ref = trainer.train(dict(params), DMatrix(X, label=y), 4, verbose_eval=False)
+        monkeypatch.delenv("SMXGB_NO_DEVICE_GROW")
+        dev = trainer.train(dict(params), DMatrix(X, label=y), 4, verbose_eval=False)
+
+        def sig(b):
+            return _json.dumps(
+                [
+                    {
+                        "f": t.feature.tolist(),
+                        "b": t.split_bin.tolist(),
+                        "l": t.left.tolist(),
+                    }
+                    for t in b.trees
+                ]
+            )
+
+        assert sig(ref) == sig(dev), "device grower diverged from per-level path"
+        p_ref = ref.predict(X[:2000])
+        p_dev = dev.predict(X[:2000])
+        np.testing.assert_allclose(p_ref, p_dev, rtol=1e-4, atol=1e-5)
+
+    def test_with_missing_and_subsample(self):
+        from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+        from sagemaker_xgboost_container_amd.models import trainer
+
+        rng = np.random.default_rng(12)
+        X = rng.normal(size=(100_000, 12)).astype(np.float32)
+        X[rng.random(size=X.shape) < 0.05] = np.nan
+        y = (np.nan_to_num(X[:, 0]) > 0).astype(np.float32)
+        res = {}
+        trainer.train(
+            {"objective": "binary:logistic", "max_depth": 5, "subsample": 0.8,
+             "colsample_bytree": 0.8, "seed": 3, "device": "cuda"},
+            DMatrix(X, label=y),
+            6,
+            evals=[(DMatrix(X, label=y), "train")],
+            evals_result=res,
+            verbose_eval=False,
+        )
+        assert res["train"]["logloss"][-1] < res["train"]["logloss"][0]
