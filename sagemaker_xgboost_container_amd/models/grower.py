@@ -244,9 +244,12 @@ class HistGrower:
         G, H = float(root_np[0]), float(root_np[1])
         root = tree.add_node(parent=-1, value=self._weight(G, H) * p.eta, sum_hess=H)
         leaf_jobs = []
-        stack = [(0, 0, root, 0, cap, G, H)]
+        from collections import deque
+
+        # BFS so node numbering matches the per-level path exactly
+        stack = deque([(0, 0, root, 0, cap, G, H)])
         while stack:
-            d, i, nid, start, end, g, h = stack.pop()
+            d, i, nid, start, end, g, h = stack.popleft()
             hidx = (1 << d) - 1 + i
             gain = float(splits_np[hidx, 0]) if d < D else -1.0
             if d < D and gain > 0.0:
