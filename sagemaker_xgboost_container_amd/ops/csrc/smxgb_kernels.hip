@@ -1195,6 +1195,121 @@ void grow_partition_level(torch::Tensor src_bins, torch::Tensor src_gh, torch::T
   }
 }
 
+// Whole-tree enqueue: the entire depthwise grow issued from C++ in one
+// Python call (single-process path; the distributed path keeps the Python
+// per-level loop so torch.distributed can enqueue the allreduce).
+void grow_tree_enqueue(
+    torch::Tensor init_bins, torch::Tensor init_gh, torch::Tensor init_rows,
+    torch::Tensor bins0, torch::Tensor gh0, torch::Tensor rows0,
+    torch::Tensor bins1, torch::Tensor gh1, torch::Tensor rows1,
+    torch::Tensor nodes, torch::Tensor node_gh, torch::Tensor splits, torch::Tensor counts,
+    torch::Tensor hist_f32, torch::Tensor acc, torch::Tensor cands,
+    std::vector<torch::Tensor> hp, std::vector<torch::Tensor> pp, torch::Tensor work,
+    torch::Tensor nbins, torch::Tensor feat_mask, torch::Tensor gh_max,
+    int64_t D, int64_t cap, int64_t nfeat, int64_t stride, int64_t n_groups,
+    int64_t feats_per_group, int64_t lds_words, int64_t has_missing, int64_t missing_bin,
+    int64_t rows_per_block, int64_t max_blocks, int64_t hist_grid, int64_t part_grid,
+    double reg_lambda, double reg_alpha, double gamma_, double min_child_weight) {
+  CHECK_GPU(init_bins);
+  auto stream = current_stream();
+  const long long slots2 = (long long)nfeat * stride * 2;
+  const bool u8 = init_bins.scalar_type() == torch::kUInt8;
+  const size_t lds_bytes = (size_t)lds_words * sizeof(unsigned long long);
+  auto mono = torch::Tensor();  // device path excludes monotone constraints
+
+  hipMemsetAsync(counts.data_ptr<int>(), 0, sizeof(int) * 2 * counts.size(0), stream);
+
+  for (int d = 0; d < (int)D; ++d) {
+    const int k = 1 << d;
+    const int base = k - 1;
+    LevelNode* nodes_d = (LevelNode*)nodes.data_ptr<int>() + base;
+    float* gh_d = node_gh.data_ptr<float>() + (long long)base * 2;
+    float* splits_d = splits.data_ptr<float>() + (long long)base * 6;
+    int* counts_d = counts.data_ptr<int>() + (long long)base * 2;
+    float* hist_d = hist_f32.data_ptr<float>() + (long long)base * slots2;
+    int* hp_d = hp[d].data_ptr<int>();
+    int* pp_d = pp[d].data_ptr<int>();
+    LevelWork* work_d = (LevelWork*)(work.data_ptr<int>() + 2 * d);
+
+    if (d == 0) {
+      hipLaunchKernelGGL(make_root_kernel, dim3(1), dim3(1), 0, stream, nodes_d, hp_d, pp_d,
+                         work_d, (int)cap, (int)rows_per_block, (int)max_blocks);
+    } else {
+      const int pk = k >> 1;
+      const int pbase = pk - 1;
+      hipLaunchKernelGGL(make_level_kernel, dim3(1), dim3(GROW_MAX_SLOTS), 0, stream,
+                         (const LevelNode*)nodes.data_ptr<int>() + pbase,
+                         splits.data_ptr<float>() + (long long)pbase * 6,
+                         counts.data_ptr<int>() + (long long)pbase * 2,
+                         node_gh.data_ptr<float>() + (long long)pbase * 2, nodes_d, gh_d, hp_d,
+                         pp_d, work_d, pk, (int)rows_per_block, (int)max_blocks);
+    }
+
+    const bool level0 = d == 0;
+    const bool par = (d % 2) == 1;
+    const void* src_bins = level0 ? init_bins.data_ptr() : (par ? bins1.data_ptr() : bins0.data_ptr());
+    const float2* src_gh = (const float2*)(level0 ? init_gh.data_ptr<float>()
+                                                  : (par ? gh1.data_ptr<float>() : gh0.data_ptr<float>()));
+    const int* src_rows = level0 ? init_rows.data_ptr<int>()
+                                 : (par ? rows1.data_ptr<int>() : rows0.data_ptr<int>());
+    const bool dpar = (d % 2) == 0;  // dst index = 1 - d%2
+    void* dst_bins = dpar ? bins1.data_ptr() : bins0.data_ptr();
+    float2* dst_gh = (float2*)(dpar ? gh1.data_ptr<float>() : gh0.data_ptr<float>());
+    int* dst_rows = dpar ? rows1.data_ptr<int>() : rows0.data_ptr<int>();
+
+    hipMemsetAsync(acc.data_ptr<int64_t>(), 0, sizeof(int64_t) * (size_t)k * slots2, stream);
+    if (u8) {
+      hipLaunchKernelGGL(hist_device_kernel<unsigned char>, dim3((int)hist_grid), dim3(HIST_BLOCK),
+                         lds_bytes, stream, (const unsigned char*)src_bins, src_gh, nodes_d, hp_d,
+                         work_d, (unsigned long long*)acc.data_ptr<int64_t>(), k, (int)nfeat,
+                         (int)stride, (int)n_groups, (int)feats_per_group,
+                         gh_max.data_ptr<float>(), (int)rows_per_block);
+    } else {
+      hipLaunchKernelGGL(hist_device_kernel<short>, dim3((int)hist_grid), dim3(HIST_BLOCK),
+                         lds_bytes, stream, (const short*)src_bins, src_gh, nodes_d, hp_d, work_d,
+                         (unsigned long long*)acc.data_ptr<int64_t>(), k, (int)nfeat, (int)stride,
+                         (int)n_groups, (int)feats_per_group, gh_max.data_ptr<float>(),
+                         (int)rows_per_block);
+    }
+
+    {
+      const long long total = (long long)k * slots2;
+      const int grid = (int)std::min<long long>((total + HIST_BLOCK - 1) / HIST_BLOCK, 2048);
+      hipLaunchKernelGGL(convert_level_kernel, dim3(std::max(grid, 1)), dim3(HIST_BLOCK), 0,
+                         stream, (const unsigned long long*)acc.data_ptr<int64_t>(), hist_d,
+                         nodes_d, k, slots2, gh_max.data_ptr<float>());
+      if (d > 0) {
+        const int pbase = (k >> 1) - 1;
+        hipLaunchKernelGGL(derive_level_kernel, dim3(std::max(grid, 1)), dim3(HIST_BLOCK), 0,
+                           stream, hist_d, hist_f32.data_ptr<float>() + (long long)pbase * slots2,
+                           nodes_d, k, slots2);
+      }
+    }
+
+    const unsigned char* mask_ptr =
+        feat_mask.numel() ? feat_mask.data_ptr<unsigned char>() : nullptr;
+    hipLaunchKernelGGL(split_scan_kernel, dim3(k * (int)nfeat), dim3(SPLIT_BLOCK), 0, stream,
+                       hist_d, (const float2*)gh_d, nbins.data_ptr<int>(), mask_ptr, nullptr,
+                       (SplitCand*)cands.data_ptr<float>(), k, (int)nfeat, (int)stride,
+                       (int)has_missing, 0, (float)reg_lambda, (float)reg_alpha, (float)gamma_,
+                       (float)min_child_weight);
+    hipLaunchKernelGGL(split_reduce_kernel, dim3(k), dim3(SPLIT_BLOCK), 0, stream,
+                       (const SplitCand*)cands.data_ptr<float>(), splits_d, k, (int)nfeat);
+
+    if (u8) {
+      hipLaunchKernelGGL(partition_device_kernel<unsigned char>, dim3((int)part_grid),
+                         dim3(HIST_BLOCK), 0, stream, (const unsigned char*)src_bins, src_gh,
+                         src_rows, (unsigned char*)dst_bins, dst_gh, dst_rows, nodes_d, pp_d,
+                         work_d, splits_d, counts_d, k, (int)nfeat, (int)missing_bin);
+    } else {
+      hipLaunchKernelGGL(partition_device_kernel<short>, dim3((int)part_grid), dim3(HIST_BLOCK),
+                         0, stream, (const short*)src_bins, src_gh, src_rows, (short*)dst_bins,
+                         dst_gh, dst_rows, nodes_d, pp_d, work_d, splits_d, counts_d, k,
+                         (int)nfeat, (int)missing_bin);
+    }
+  }
+}
+
 void find_splits(torch::Tensor hist, torch::Tensor parent, torch::Tensor nbins,
                  torch::Tensor feat_mask, torch::Tensor monotone, torch::Tensor cands,
                  torch::Tensor out, int64_t k, int64_t f, int64_t stride, int64_t has_missing,
@@ -1233,6 +1348,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grow_convert_level", &grow_convert_level);
   m.def("grow_derive_level", &grow_derive_level);
   m.def("grow_partition_level", &grow_partition_level);
+  m.def("grow_tree_enqueue", &grow_tree_enqueue, "whole depthwise tree enqueued from one call");
   m.def("partition_compact", &partition_compact, "compacting partition (rows+bins+gh rewrite)");
   m.def("leaf_update_compact", &leaf_update_compact, "leaf scatter from compact row ids");
   m.def("leaf_update", &leaf_update, "batched leaf value scatter into margins");

@@ -526,6 +526,25 @@ class DeviceGrower:
             comm.allreduce_(root_gh)
         self.node_gh[0] = root_gh.to(torch.float32)
 
+        if comm is None:
+            # whole tree enqueued from ONE extension call
+            _K.grow_tree_enqueue(
+                st._bins_init, st._gh_init, st._rows_init,
+                st._bins[0], st._gh[0], st._rows[0],
+                st._bins[1], st._gh[1], st._rows[1],
+                self.nodes, self.node_gh, self.splits, self.counts,
+                self.hist_f32, self.acc, self.cands, self.hp, self.pp, self.work,
+                qm._nbins_i32, self.mask, scale,
+                self.D, st.cap, f, stride, self.n_groups, self.feats_per_group,
+                self.lds_words, 1 if qm.has_missing else 0, missing_bin,
+                _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB, _GROW_HIST_GRID, _GROW_PART_GRID,
+                reg_lambda, reg_alpha, gamma, mcw,
+            )
+            splits_np = self.splits.cpu().numpy()
+            counts_np = self.counts.cpu().numpy()
+            root_np = self.node_gh[0].cpu().numpy()
+            return splits_np, counts_np, root_np
+
         for d in range(self.D):
             k = 1 << d
             base = k - 1
