@@ -33,8 +33,10 @@ import os as _os
 # LDS histogram slab per feature group (56 KiB -> 2 blocks/CU co-residency).
 # Env-tunable for measured sweeps (SMXGB_LDS_KB / SMXGB_ROWS_PER_BLOCK).
 _LDS_BYTES = int(_os.environ.get("SMXGB_LDS_KB", "56")) * 1024
-_ROWS_PER_BLOCK = int(_os.environ.get("SMXGB_ROWS_PER_BLOCK", "8192"))
-_MAX_BLOCKS_PER_JOB = int(_os.environ.get("SMXGB_MAX_BLOCKS", "512"))
+# measured sweep (profiles/r01_optimization_log.md): 4096-row blocks with a
+# 2048 cap keep every CU fed at the deep levels
+_ROWS_PER_BLOCK = int(_os.environ.get("SMXGB_ROWS_PER_BLOCK", "4096"))
+_MAX_BLOCKS_PER_JOB = int(_os.environ.get("SMXGB_MAX_BLOCKS", "2048"))
 
 
 def compute_scale(gh, comm=None):
@@ -461,8 +463,8 @@ class _V1TreeState:
 
 
 
-_GROW_HIST_GRID = int(_os.environ.get("SMXGB_GROW_HIST_GRID", "768"))
-_GROW_PART_GRID = int(_os.environ.get("SMXGB_GROW_PART_GRID", "1024"))
+_GROW_HIST_GRID = int(_os.environ.get("SMXGB_GROW_HIST_GRID", "1536"))
+_GROW_PART_GRID = int(_os.environ.get("SMXGB_GROW_PART_GRID", "4096"))
 
 
 class DeviceGrower:
