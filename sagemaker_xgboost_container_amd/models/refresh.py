@@ -144,8 +144,12 @@ def run_update_process(booster, params, X, y, weight, objective, margin, n_outpu
 
         results = []
         for es in eval_sets:
-            es.margin = booster._margin(es.X).reshape(es.X.shape[0], -1).clone()
-            m = es.margin.squeeze(1) if n_outputs == 1 else es.margin
+            if es.is_train:
+                es_margin = margin
+            else:
+                es.margin = booster._margin(es.X).reshape(es.X.shape[0], -1).clone()
+                es_margin = es.margin
+            m = es_margin.squeeze(1) if n_outputs == 1 else es_margin
             for metric_name in metric_names:
                 results.append((es.name, metric_name, evaluate_metric(metric_name, m, es.y, es.w, objective)))
         if container.after_iteration(booster, epoch, results):

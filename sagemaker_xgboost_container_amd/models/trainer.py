@@ -51,10 +51,24 @@ def _resolve_device(params):
 
 
 class _EvalSet:
-    """Resident eval data: dense features + incrementally updated margins."""
+    """Resident eval data: dense features + incrementally updated margins.
 
-    def __init__(self, dmatrix, name, device, n_outputs, base_margin_value):
+    When the eval set IS the training DMatrix (the standard watchlist), it
+    aliases the training tensors and margins — no copy, no per-round
+    traversal (`is_train`).
+    """
+
+    def __init__(self, dmatrix, name, device, n_outputs, base_margin_value,
+                 train_dmatrix=None, train_X=None, train_y=None, train_w=None):
         self.name = name
+        self.is_train = dmatrix is train_dmatrix
+        if self.is_train:
+            self.X = train_X
+            self.y = train_y
+            self.w = train_w
+            self.dmatrix = dmatrix
+            self.margin = None  # provided by the trainer each round
+            return
         self.X = torch.as_tensor(dmatrix.to_dense(), dtype=torch.float32, device=device)
         self.y = torch.as_tensor(dmatrix.get_label(), dtype=torch.float32, device=device)
         w = dmatrix.get_weight()
@@ -151,11 +165,14 @@ def train(
             margin[:, booster.tree_info[t_idx]] += backend.predict_tree(tree, X)
 
     eval_sets = [
-        _EvalSet(dm, name, device, n_outputs, base_margin_value) for dm, name in (evals or []) if dm is not None
+        _EvalSet(dm, name, device, n_outputs, base_margin_value,
+                 train_dmatrix=dtrain, train_X=X, train_y=y, train_w=weight)
+        for dm, name in (evals or [])
+        if dm is not None
     ]
     # re-play existing trees into eval margins on warm start
     for es in eval_sets:
-        if es.dmatrix.get_base_margin() is None:
+        if not es.is_train and es.dmatrix.get_base_margin() is None:
             for t_idx, tree in enumerate(booster.trees):
                 es.margin[:, booster.tree_info[t_idx]] += backend.predict_tree(tree, es.X)
 
@@ -253,11 +270,12 @@ def train(
                 margin[:, cls] += delta
             booster.add_iteration([], [])
             for es in eval_sets:
-                es.margin = (
-                    torch.nan_to_num(es.X, nan=0.0) @ booster.linear_model.weights
-                    + booster.linear_model.bias
-                    + float(base_margin_value)
-                )
+                if not es.is_train:
+                    es.margin = (
+                        torch.nan_to_num(es.X, nan=0.0) @ booster.linear_model.weights
+                        + booster.linear_model.bias
+                        + float(base_margin_value)
+                    )
         else:
             # dart scaling factors for this round's new trees
             k_drop = len(dropped)
@@ -285,7 +303,8 @@ def train(
                     round_trees.append(tree)
                     round_info.append(cls)
                     for es in eval_sets:
-                        es.margin[:, cls] += backend.predict_tree(tree, es.X) * new_tree_scale
+                        if not es.is_train:
+                            es.margin[:, cls] += backend.predict_tree(tree, es.X) * new_tree_scale
             # rescale dropped trees and restore their (scaled) contribution
             for t in dropped:
                 booster.weight_drop[t] *= dropped_factor
@@ -295,12 +314,14 @@ def train(
                 # dropped-tree rescaling invalidates the incremental eval
                 # margins: recompute them from the booster exactly
                 for es in eval_sets:
-                    es.margin = booster._margin(es.X).reshape(es.X.shape[0], -1).clone()
+                    if not es.is_train:
+                        es.margin = booster._margin(es.X).reshape(es.X.shape[0], -1).clone()
 
         # -- evaluation ----------------------------------------------------
         results = []
         for es in eval_sets:
-            m = es.margin.squeeze(1) if n_outputs == 1 else es.margin
+            es_margin = margin if es.is_train else es.margin
+            m = es_margin.squeeze(1) if n_outputs == 1 else es_margin
             for metric_name in metric_names:
                 value = evaluate_metric(metric_name, m, es.y, es.w, objective)
                 results.append((es.name, metric_name, value))
