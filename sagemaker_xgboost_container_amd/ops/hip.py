@@ -48,8 +48,14 @@ def compute_scale(gh, comm=None):
     return m
 
 
+def _padded_words(pairs):
+    """LDS u64 words for `pairs` histogram slots incl. bank-skew padding."""
+    return (pairs + (pairs >> 3) + 1) * 2
+
+
 def _feature_groups(nfeat, stride):
-    per_group = max(1, min(nfeat, (_LDS_BYTES // 16) // max(stride, 1)))
+    # 18 bytes per slot effective (16 + 1/8 padding)
+    per_group = max(1, min(nfeat, (_LDS_BYTES // 18) // max(stride, 1)))
     groups = []
     f = 0
     while f < nfeat:
@@ -110,7 +116,7 @@ def build_histograms(qm, gh, rowbuf, jobs, scale):
 
     jobs_dev = _pack_jobs(packed)
     block_job = _block_map(blocks_per)
-    lds_words = max((fe - fs) for fs, fe in groups) * stride * 2
+    lds_words = _padded_words(max((fe - fs) for fs, fe in groups) * stride)
     if isinstance(scale, torch.Tensor):  # device-resident (gmax, hmax)
         gmax, hmax = (float(v) for v in scale.cpu())
         scale = (2.0**_FIXED_BITS / max(gmax, 1e-30), 2.0**_FIXED_BITS / max(hmax, 1e-30))
@@ -360,7 +366,7 @@ class TreeState:
                 blocks_per.append(nb)
                 first_block += nb
         jobs_dev, block_job = _pack_jobs_and_map(packed, blocks_per)
-        lds_words = max((fe - fs) for fs, fe in groups) * stride * 2
+        lds_words = _padded_words(max((fe - fs) for fs, fe in groups) * stride)
         _K.hist_build_compact(
             bins_c, gh_c, jobs_dev, block_job, acc, f, stride, scale, lds_words
         )
@@ -503,7 +509,7 @@ class DeviceGrower:
         groups = _feature_groups(f, stride)
         self.n_groups = len(groups)
         self.feats_per_group = groups[0][1] - groups[0][0]
-        self.lds_words = max(fe - fs for fs, fe in groups) * stride * 2
+        self.lds_words = _padded_words(max(fe - fs for fs, fe in groups) * stride)
         if feature_mask is None:
             self.mask = torch.empty(0, dtype=torch.uint8, device=device)
         else:
