@@ -24,14 +24,18 @@
 #define WAVE 64
 #define HIST_BLOCK 256
 
-// LDS histogram bank-conflict padding: a 16-B (u64 pair) slot at linear
-// index s occupies banks 4s..4s+3 (mod 32) — only 8 distinct bank groups
-// for any run of bins, measured 75% of LDS cycles lost to conflicts
-// (SQ_LDS_BANK_CONFLICT 0.74G vs SQ_LDS_IDX_ACTIVE 1.0G). Skewing by s/8
-// rotates each 8-slot group across all 32 banks.
+// LDS histogram layout: separate g and h arrays (SoA). A 16-B interleaved
+// (g,h) slot can only start on 8 of the 32 bank groups (measured 75% of
+// LDS cycles lost to conflicts: SQ_LDS_BANK_CONFLICT 0.74G vs
+// SQ_LDS_IDX_ACTIVE 1.0G); 8-B-strided u64 arrays reach 16 groups, halving
+// intrinsic conflict pressure. lds_pad_slot adds a skew word per 8 slots.
 __device__ __host__ inline int lds_pad_slot(int s) { return s + (s >> 3); }
 __device__ __host__ inline long long lds_padded_words(long long pairs) {
+  // per array: pairs + pairs/8 + 1 u64 words; x2 for the g and h arrays
   return (pairs + (pairs >> 3) + 1) * 2;
+}
+__device__ inline long long lds_half_words(long long pairs) {
+  return pairs + (pairs >> 3) + 1;
 }
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be a ROCm device tensor")
 
@@ -83,7 +87,8 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_kernel(
   const HistJob job = jobs[block_job[blockIdx.x]];
   const int nf_group = job.fg_end - job.fg_start;
   const int lds_words = nf_group * stride * 2;
-  const int lds_padded = (int)lds_padded_words(nf_group * stride);
+  const int hofs = (int)lds_half_words(nf_group * stride);
+  const int lds_padded = 2 * hofs;
   for (int i = threadIdx.x; i < lds_padded; i += blockDim.x) lhist[i] = 0ull;
   __syncthreads();
 
@@ -104,27 +109,27 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_kernel(
         for (int f4 = 0; f4 < (nf_group >> 2); ++f4) {
           const uchar4 b4 = rp4[f4];
           const int base = (f4 << 2) * stride;
-          const int s0 = lds_pad_slot(base + (int)b4.x) * 2;
-          const int s1 = lds_pad_slot(base + stride + (int)b4.y) * 2;
-          const int s2 = lds_pad_slot(base + 2 * stride + (int)b4.z) * 2;
-          const int s3 = lds_pad_slot(base + 3 * stride + (int)b4.w) * 2;
+          const int s0 = lds_pad_slot(base + (int)b4.x);
+          const int s1 = lds_pad_slot(base + stride + (int)b4.y);
+          const int s2 = lds_pad_slot(base + 2 * stride + (int)b4.z);
+          const int s3 = lds_pad_slot(base + 3 * stride + (int)b4.w);
           atomicAdd(&lhist[s0], gfix);
-          atomicAdd(&lhist[s0 + 1], hfix);
+          atomicAdd(&lhist[hofs + s0], hfix);
           atomicAdd(&lhist[s1], gfix);
-          atomicAdd(&lhist[s1 + 1], hfix);
+          atomicAdd(&lhist[hofs + s1], hfix);
           atomicAdd(&lhist[s2], gfix);
-          atomicAdd(&lhist[s2 + 1], hfix);
+          atomicAdd(&lhist[hofs + s2], hfix);
           atomicAdd(&lhist[s3], gfix);
-          atomicAdd(&lhist[s3 + 1], hfix);
+          atomicAdd(&lhist[hofs + s3], hfix);
         }
         continue;
       }
     }
     #pragma unroll 4
     for (int f = 0; f < nf_group; ++f) {
-      const int slot = lds_pad_slot(f * stride + (int)rp[f]) * 2;
+      const int slot = lds_pad_slot(f * stride + (int)rp[f]);
       atomicAdd(&lhist[slot], gfix);
-      atomicAdd(&lhist[slot + 1], hfix);
+      atomicAdd(&lhist[hofs + slot], hfix);
     }
   }
   __syncthreads();
@@ -133,7 +138,8 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_kernel(
       out + ((long long)job.hist_idx * nfeat + job.fg_start) * (long long)stride * 2;
   for (int i = threadIdx.x; i < lds_words; i += blockDim.x) {
     const int pair = i >> 1;
-    const unsigned long long v = lhist[lds_pad_slot(pair) * 2 + (i & 1)];
+    const int idx = lds_pad_slot(pair) + ((i & 1) ? hofs : 0);
+    const unsigned long long v = lhist[idx];
     if (v) atomicAdd(&gout[i], v);
   }
 }
@@ -189,7 +195,8 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_compact_kernel(
   const HistJob job = jobs[block_job[blockIdx.x]];
   const int nf_group = job.fg_end - job.fg_start;
   const int lds_words = nf_group * stride * 2;
-  const int lds_padded = (int)lds_padded_words(nf_group * stride);
+  const int hofs = (int)lds_half_words(nf_group * stride);
+  const int lds_padded = 2 * hofs;
   for (int i = threadIdx.x; i < lds_padded; i += blockDim.x) lhist[i] = 0ull;
   __syncthreads();
 
@@ -207,27 +214,27 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_compact_kernel(
         for (int f4 = 0; f4 < (nf_group >> 2); ++f4) {
           const uchar4 b4 = rp4[f4];
           const int base = (f4 << 2) * stride;
-          const int s0 = lds_pad_slot(base + (int)b4.x) * 2;
-          const int s1 = lds_pad_slot(base + stride + (int)b4.y) * 2;
-          const int s2 = lds_pad_slot(base + 2 * stride + (int)b4.z) * 2;
-          const int s3 = lds_pad_slot(base + 3 * stride + (int)b4.w) * 2;
+          const int s0 = lds_pad_slot(base + (int)b4.x);
+          const int s1 = lds_pad_slot(base + stride + (int)b4.y);
+          const int s2 = lds_pad_slot(base + 2 * stride + (int)b4.z);
+          const int s3 = lds_pad_slot(base + 3 * stride + (int)b4.w);
           atomicAdd(&lhist[s0], gfix);
-          atomicAdd(&lhist[s0 + 1], hfix);
+          atomicAdd(&lhist[hofs + s0], hfix);
           atomicAdd(&lhist[s1], gfix);
-          atomicAdd(&lhist[s1 + 1], hfix);
+          atomicAdd(&lhist[hofs + s1], hfix);
           atomicAdd(&lhist[s2], gfix);
-          atomicAdd(&lhist[s2 + 1], hfix);
+          atomicAdd(&lhist[hofs + s2], hfix);
           atomicAdd(&lhist[s3], gfix);
-          atomicAdd(&lhist[s3 + 1], hfix);
+          atomicAdd(&lhist[hofs + s3], hfix);
         }
         continue;
       }
     }
     #pragma unroll 4
     for (int f = 0; f < nf_group; ++f) {
-      const int slot = lds_pad_slot(f * stride + (int)rp[f]) * 2;
+      const int slot = lds_pad_slot(f * stride + (int)rp[f]);
       atomicAdd(&lhist[slot], gfix);
-      atomicAdd(&lhist[slot + 1], hfix);
+      atomicAdd(&lhist[hofs + slot], hfix);
     }
   }
   __syncthreads();
@@ -236,7 +243,8 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_compact_kernel(
       out + ((long long)job.hist_idx * nfeat + job.fg_start) * (long long)stride * 2;
   for (int i = threadIdx.x; i < lds_words; i += blockDim.x) {
     const int pair = i >> 1;
-    const unsigned long long v = lhist[lds_pad_slot(pair) * 2 + (i & 1)];
+    const int idx = lds_pad_slot(pair) + ((i & 1) ? hofs : 0);
+    const unsigned long long v = lhist[idx];
     if (v) atomicAdd(&gout[i], v);
   }
 }
@@ -512,7 +520,8 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_device_kernel(
     const int fg_start = fg * feats_per_group;
     const int nf_group = min(feats_per_group, nfeat - fg_start);
     const int lds_words = nf_group * stride * 2;
-    const int lds_padded = (int)lds_padded_words(nf_group * stride);
+    const int hofs = (int)lds_half_words(nf_group * stride);
+    const int lds_padded = 2 * hofs;
 
     for (int i = threadIdx.x; i < lds_padded; i += blockDim.x) lhist[i] = 0ull;
     __syncthreads();
@@ -526,9 +535,9 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_device_kernel(
       const BinT* rp = bins_c + (long long)r * nfeat + fg_start;
       #pragma unroll 4
       for (int f = 0; f < nf_group; ++f) {
-        const int slot2 = lds_pad_slot(f * stride + (int)rp[f]) * 2;
+        const int slot2 = lds_pad_slot(f * stride + (int)rp[f]);
         atomicAdd(&lhist[slot2], gfix);
-        atomicAdd(&lhist[slot2 + 1], hfix);
+        atomicAdd(&lhist[hofs + slot2], hfix);
       }
     }
     __syncthreads();
@@ -536,7 +545,8 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_device_kernel(
         out + ((long long)slot * nfeat + fg_start) * (long long)stride * 2;
     for (int i = threadIdx.x; i < lds_words; i += blockDim.x) {
       const int pair = i >> 1;
-      const unsigned long long v = lhist[lds_pad_slot(pair) * 2 + (i & 1)];
+      const int idx = lds_pad_slot(pair) + ((i & 1) ? hofs : 0);
+      const unsigned long long v = lhist[idx];
       if (v) atomicAdd(&gout[i], v);
     }
     __syncthreads();
