@@ -135,6 +135,7 @@ class Booster:
         iteration_range=None,
         validate_features=True,
         pred_contribs=False,
+        pred_leaf=False,
         training=False,
         ntree_limit=None,
     ):
@@ -144,6 +145,8 @@ class Booster:
             iteration_range = (0, int(ntree_limit) // max(1, self._trees_per_round()))
         if pred_contribs:
             return self._pred_contribs(X)
+        if pred_leaf:
+            return self._pred_leaf(X)
         margin = self._margin(X, iteration_range)
         if output_margin:
             return margin.cpu().numpy()
@@ -176,6 +179,89 @@ class Booster:
                 )
         device = "cuda" if torch.cuda.is_available() and self.params.get("predictor") != "cpu_predictor" else "cpu"
         return torch.as_tensor(arr, dtype=torch.float32, device=device)
+
+    def _pred_leaf(self, X):
+        """Leaf indices per (row, tree) — xgboost pred_leaf=True."""
+        Xc = X.cpu().numpy()
+        n = Xc.shape[0]
+        out = np.zeros((n, len(self.trees)), dtype=np.int32)
+        for t, tree in enumerate(self.trees):
+            for i in range(n):
+                nid = 0
+                while tree.left[nid] >= 0:
+                    fv = Xc[i, tree.feature[nid]]
+                    if np.isnan(fv):
+                        nid = tree.left[nid] if tree.default_left[nid] else tree.right[nid]
+                    else:
+                        nid = tree.left[nid] if fv < tree.threshold[nid] else tree.right[nid]
+                out[i, t] = nid
+        return out
+
+    def get_score(self, fmap="", importance_type="weight"):
+        """Feature importances (weight / gain / total_gain / cover /
+        total_cover) keyed like xgboost ('f<idx>' or feature names)."""
+        counts = {}
+        gains = {}
+        covers = {}
+        for tree in self.trees:
+            for nid in range(tree.num_nodes):
+                if tree.left[nid] < 0:
+                    continue
+                f = int(tree.feature[nid])
+                counts[f] = counts.get(f, 0) + 1
+                gains[f] = gains.get(f, 0.0) + float(tree.gain[nid])
+                covers[f] = covers.get(f, 0.0) + float(tree.sum_hess[nid])
+
+        def name(f):
+            if self.feature_names and f < len(self.feature_names):
+                return self.feature_names[f]
+            return f"f{f}"
+
+        if importance_type == "weight":
+            return {name(f): v for f, v in counts.items()}
+        if importance_type == "total_gain":
+            return {name(f): v for f, v in gains.items()}
+        if importance_type == "gain":
+            return {name(f): gains[f] / counts[f] for f in counts}
+        if importance_type == "total_cover":
+            return {name(f): v for f, v in covers.items()}
+        if importance_type == "cover":
+            return {name(f): covers[f] / counts[f] for f in counts}
+        raise ValueError(f"Unknown importance type: {importance_type}")
+
+    def get_dump(self, fmap="", with_stats=False, dump_format="text"):
+        """Per-tree text dumps in the xgboost format."""
+        dumps = []
+        for tree in self.trees:
+            lines = []
+
+            def walk(nid, depth):
+                indent = "\t" * depth
+                if tree.left[nid] < 0:
+                    stat = f",cover={tree.sum_hess[nid]}" if with_stats else ""
+                    lines.append(f"{indent}{nid}:leaf={tree.value[nid]}{stat}")
+                else:
+                    f = int(tree.feature[nid])
+                    fname = (
+                        self.feature_names[f]
+                        if self.feature_names and f < len(self.feature_names)
+                        else f"f{f}"
+                    )
+                    yes, no = int(tree.left[nid]), int(tree.right[nid])
+                    missing = yes if tree.default_left[nid] else no
+                    stat = (
+                        f",gain={tree.gain[nid]},cover={tree.sum_hess[nid]}" if with_stats else ""
+                    )
+                    lines.append(
+                        f"{indent}{nid}:[{fname}<{tree.threshold[nid]}] "
+                        f"yes={yes},no={no},missing={missing}{stat}"
+                    )
+                    walk(yes, depth + 1)
+                    walk(no, depth + 1)
+
+            walk(0, 0)
+            dumps.append("\n".join(lines) + "\n")
+        return dumps
 
     def _pred_contribs(self, X):
         """TreeSHAP-style contributions (approximate: Saabas method), host."""

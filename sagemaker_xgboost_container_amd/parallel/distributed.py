@@ -100,13 +100,26 @@ class Rabit:
             "Connecting to cluster master %s:%s as rank %d/%d",
             master_ip, self.port, self.rank, self.n_workers,
         )
-        dist.init_process_group(
-            backend=self.backend,
-            init_method=f"tcp://{master_ip}:{self.port}",
-            rank=self.rank,
-            world_size=self.n_workers,
-            timeout=datetime.timedelta(seconds=self.timeout_s),
-        )
+        # startup robustness: retry the rendezvous (reference retries
+        # CommunicatorContext init, distributed.py:215-227)
+        last_error = None
+        for attempt in range(3):
+            try:
+                dist.init_process_group(
+                    backend=self.backend,
+                    init_method=f"tcp://{master_ip}:{self.port}",
+                    rank=self.rank,
+                    world_size=self.n_workers,
+                    timeout=datetime.timedelta(seconds=self.timeout_s),
+                )
+                last_error = None
+                break
+            except Exception as e:  # noqa: BLE001 - connection errors vary by backend
+                last_error = e
+                logger.warning("Cluster rendezvous attempt %d failed: %s", attempt + 1, e)
+                time.sleep(2 * (attempt + 1))
+        if last_error is not None:
+            raise exc.PlatformError("Could not join the training cluster", caused_by=last_error)
         self.comm = comm_mod.Communicator()
         return RabitHelper(self.rank == 0, self.current_host, self.port, self.comm)
 
