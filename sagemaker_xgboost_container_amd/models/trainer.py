@@ -111,6 +111,13 @@ def train(
     for key in unknown:
         logger.warning("Ignoring unknown training parameter: %s", key)
 
+    verbosity = int(params.get("verbosity", 1) or 1)
+    logging.getLogger("sagemaker_xgboost_container_amd").setLevel(
+        {0: logging.ERROR, 1: logging.WARNING, 2: logging.INFO, 3: logging.DEBUG}.get(
+            verbosity, logging.WARNING
+        )
+    )
+
     device = _resolve_device(params)
     seed = int(params.get("seed", 0) or 0)
     generator = torch.Generator(device=device)
