@@ -43,14 +43,26 @@ def _list_data_files(path):
     return files
 
 
-def _parse_csv_files(files, delimiter=",", label_column=0, weight_column=None):
-    import pandas as pd
+def _parse_csv_native(files, delimiter):
+    import torch  # noqa: F401
 
-    frames = [
-        pd.read_csv(f, sep=delimiter, header=None, dtype=np.float32, na_values=[""], skip_blank_lines=True)
-        for f in files
-    ]
-    data = pd.concat(frames, axis=0, ignore_index=True).to_numpy(dtype=np.float32)
+    from ..ops import _smxgb_hip as K
+
+    text = "".join(open(f, "r", errors="ignore").read() for f in files)
+    return K.parse_csv(text, delimiter, 0).numpy()
+
+
+def _parse_csv_files(files, delimiter=",", label_column=0, weight_column=None):
+    try:
+        data = _parse_csv_native(files, delimiter)
+    except ImportError:
+        import pandas as pd
+
+        frames = [
+            pd.read_csv(f, sep=delimiter, header=None, dtype=np.float32, na_values=[""], skip_blank_lines=True)
+            for f in files
+        ]
+        data = pd.concat(frames, axis=0, ignore_index=True).to_numpy(dtype=np.float32)
     ncol = data.shape[1]
     label = None
     weight = None

@@ -111,10 +111,12 @@ def train(
     for key in unknown:
         logger.warning("Ignoring unknown training parameter: %s", key)
 
+    # verbosity 0 silent(ish), 1 default (eval lines are INFO — the
+    # CloudWatch scrape contract — so the default keeps INFO), 3 debug
     verbosity = int(params.get("verbosity", 1) or 1)
     logging.getLogger("sagemaker_xgboost_container_amd").setLevel(
-        {0: logging.ERROR, 1: logging.WARNING, 2: logging.INFO, 3: logging.DEBUG}.get(
-            verbosity, logging.WARNING
+        {0: logging.WARNING, 1: logging.INFO, 2: logging.INFO, 3: logging.DEBUG}.get(
+            verbosity, logging.INFO
         )
     )
 

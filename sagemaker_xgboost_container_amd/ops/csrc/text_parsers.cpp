@@ -12,6 +12,7 @@
 #include <atomic>
 #include <cstdlib>
 #include <cstring>
+#include <limits>
 #include <thread>
 #include <vector>
 
@@ -192,8 +193,108 @@ void predict_forest_cpu(torch::Tensor X, torch::Tensor left, torch::Tensor right
   });
 }
 
+// Multi-threaded CSV parser: numeric fields, empty -> NaN, arbitrary
+// single-char delimiter. Two passes: count rows/cols, then parse into a
+// dense float32 tensor in parallel.
+torch::Tensor parse_csv(const std::string& text, const std::string& delimiter, int64_t nthreads) {
+  const char delim = delimiter.empty() ? ',' : delimiter[0];
+  const char* data = text.data();
+  const int64_t size = (int64_t)text.size();
+
+  // column count from the first non-empty line
+  int64_t ncol = 0;
+  {
+    int64_t p = 0;
+    while (p < size && (data[p] == '\n' || data[p] == '\r')) ++p;
+    int64_t cols = 1;
+    while (p < size && data[p] != '\n') {
+      if (data[p] == delim) ++cols;
+      ++p;
+    }
+    ncol = cols;
+  }
+  // row count
+  int64_t nrow = 0;
+  {
+    bool in_line = false;
+    for (int64_t p = 0; p < size; ++p) {
+      if (data[p] == '\n') {
+        if (in_line) ++nrow;
+        in_line = false;
+      } else if (data[p] != '\r') {
+        in_line = true;
+      }
+    }
+    if (in_line) ++nrow;
+  }
+
+  auto out = torch::empty({nrow, ncol}, torch::kFloat32);
+  float* op = out.data_ptr<float>();
+
+  int64_t nt = nthreads > 0 ? nthreads : (int64_t)std::thread::hardware_concurrency();
+  if (nt < 1) nt = 1;
+  if (size < (1 << 16)) nt = 1;
+
+  // chunk by line boundaries; each chunk counts its starting row first
+  std::vector<int64_t> bounds(nt + 1, size);
+  bounds[0] = 0;
+  for (int64_t t = 1; t < nt; ++t) {
+    int64_t p = size * t / nt;
+    while (p < size && data[p] != '\n') ++p;
+    bounds[t] = p < size ? p + 1 : size;
+  }
+  std::vector<int64_t> start_row(nt + 1, 0);
+  {
+    int64_t row = 0;
+    int64_t t = 1;
+    bool in_line = false;
+    for (int64_t p = 0; p < size && t <= nt; ++p) {
+      while (t <= nt && p == bounds[t]) start_row[t++] = row;
+      if (data[p] == '\n') {
+        if (in_line) ++row;
+        in_line = false;
+      } else if (data[p] != '\r') {
+        in_line = true;
+      }
+    }
+  }
+
+  const float nan_v = std::numeric_limits<float>::quiet_NaN();
+  std::vector<std::thread> threads;
+  for (int64_t t = 0; t < nt; ++t) {
+    threads.emplace_back([&, t] {
+      const char* p = data + bounds[t];
+      const char* end = data + bounds[t + 1];
+      int64_t row = start_row[t];
+      while (p < end) {
+        while (p < end && (*p == '\n' || *p == '\r')) ++p;
+        if (p >= end) break;
+        float* rp = op + row * ncol;
+        for (int64_t c = 0; c < ncol; ++c) {
+          // empty field -> NaN
+          if (p >= end || *p == delim || *p == '\n' || *p == '\r') {
+            rp[c] = nan_v;
+          } else {
+            char* next = nullptr;
+            rp[c] = std::strtof(p, &next);
+            p = (next == p) ? p : next;
+            while (p < end && *p != delim && *p != '\n') ++p;  // trailing junk
+          }
+          if (p < end && *p == delim) ++p;
+        }
+        while (p < end && *p != '\n') ++p;
+        ++row;
+      }
+    });
+  }
+  for (auto& th : threads) th.join();
+  return out;
+}
+
 void init_text_parsers(pybind11::module_& m) {
   m.def("parse_libsvm", &parse_libsvm, "multi-threaded libsvm -> CSR parser",
         pybind11::arg("text"), pybind11::arg("nthreads") = 0);
   m.def("predict_forest_cpu", &predict_forest_cpu, "parallel host forest traversal");
+  m.def("parse_csv", &parse_csv, "multi-threaded csv -> dense float32 parser",
+        pybind11::arg("text"), pybind11::arg("delimiter") = ",", pybind11::arg("nthreads") = 0);
 }
