@@ -18,10 +18,6 @@ from .recordio_protobuf import read_recordio_protobuf
 _MIME_CSV = "text/csv"
 
 
-def _clean_csv_string(csv_string, delimiter):
-    return ["nan" if x == "" else x for x in csv_string.split(delimiter)]
-
-
 def csv_to_dmatrix(input, dtype=None):
     """CSV payload (no label column) -> DMatrix. Empty fields become NaN.
 
