@@ -252,7 +252,7 @@ __global__ __launch_bounds__(HIST_BLOCK) void hist_compact_kernel(
 // compact partition: decide per row from src bins (sequential read), then
 // copy the whole record (row id + bin bytes + gh) to its contiguous
 // destination run. Destinations inside one tile form two coalesced runs.
-#define CPART_TILE 4096
+#define CPART_TILE 2048
 
 // jobs carry segments only; the split decision (gain, feature, bin,
 // missing direction) is read from the on-device packed split tensor
