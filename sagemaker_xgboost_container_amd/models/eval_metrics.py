@@ -213,6 +213,37 @@ def interval_regression_accuracy(pred, lower, upper):
 
 METRIC_NEEDS_PROB = {"logloss", "error", "auc", "aucpr", "merror", "mlogloss"}
 
+# Distributed aggregation: every built-in metric is either a ratio-of-sums
+# over rows (weighted mean of per-rank values by per-rank weight mass is
+# EXACT), a sqrt of one (rmse/rmsle: aggregate in squared space), or a mean
+# over groups/events (ndcg/map/cox: weight by group/event count; auc/aucpr:
+# weighted average of per-worker values — xgboost's documented distributed
+# AUC semantics).
+_SQRT_METRICS = {"rmse", "rmsle"}
+_GROUP_METRICS = {"ndcg", "map"}
+
+
+def metric_mass(metric_name, y, weight, objective):
+    """The aggregation weight for one rank's metric value (see above)."""
+    base = metric_name.partition("@")[0]
+    if base in _GROUP_METRICS:
+        return float(len(_group_bounds(objective, y.shape[0], y.device))) if objective else 1.0
+    if base == "cox-nloglik":
+        return max(float((y > 0).sum()), 1.0)
+    if y.numel() == 0:
+        return 0.0
+    w = _w(weight, y)
+    return float(w.sum())
+
+
+def to_agg_space(metric_name, value):
+    """Map a metric value into the space where weighted means are exact."""
+    return value * value if metric_name.partition("@")[0] in _SQRT_METRICS else value
+
+
+def from_agg_space(metric_name, value):
+    return math.sqrt(max(value, 0.0)) if metric_name.partition("@")[0] in _SQRT_METRICS else value
+
 
 def evaluate(metric_name, margin, y, weight, objective):
     """Compute one named metric from raw margins via the objective transform."""

@@ -19,6 +19,7 @@ import torch
 
 from .booster import Booster
 from .callback_api import CallbackContainer, EarlyStopping, EvaluationMonitor
+from . import eval_metrics
 from .eval_metrics import evaluate as evaluate_metric
 from .grower import HistGrower
 from .objectives import create_objective
@@ -327,13 +328,19 @@ def train(
                         es.margin = booster._margin(es.X).reshape(es.X.shape[0], -1).clone()
 
         # -- evaluation ----------------------------------------------------
+        # Each rank evaluates its local shard; values are then aggregated to
+        # one identical number on every rank (ONE fused allreduce — exact for
+        # ratio-of-sums metrics, per-worker weighted average for rank metrics;
+        # the reference gets the same property from rabit inside xgboost).
         results = []
+        masses = []
         for es in eval_sets:
             es_margin = margin if es.is_train else es.margin
             m = es_margin.squeeze(1) if n_outputs == 1 else es_margin
             for metric_name in metric_names:
                 value = evaluate_metric(metric_name, m, es.y, es.w, objective)
                 results.append((es.name, metric_name, value))
+                masses.append(eval_metrics.metric_mass(metric_name, es.y, es.w, objective))
             if feval is not None:
                 m_np = m.cpu().numpy()
                 custom = feval(m_np, es.dmatrix)
@@ -341,6 +348,21 @@ def train(
                     custom = [custom]
                 for metric_name, value in custom:
                     results.append((es.name, metric_name, float(value)))
+                    masses.append(float(es.y.numel()))
+
+        if comm is not None and comm.world_size > 1 and results:
+            vm = torch.tensor(
+                [eval_metrics.to_agg_space(nm, v) * mass for (_, nm, v), mass in zip(results, masses)]
+                + masses,
+                dtype=torch.float64,
+                device=device,
+            )
+            comm.allreduce_(vm)
+            k = len(results)
+            results = [
+                (ds, nm, eval_metrics.from_agg_space(nm, float(vm[i] / vm[k + i]) if float(vm[k + i]) > 0 else 0.0))
+                for i, (ds, nm, _) in enumerate(results)
+            ]
 
         if container.after_iteration(booster, epoch, results):
             break

@@ -188,3 +188,75 @@ def test_sharded_matches_single_node():
         p.join(timeout=120)
         assert p.exitcode == 0
     assert np.corrcoef(p_single, p_dist)[0, 1] > 0.98
+
+
+def _eval_agg_worker(rank, world, port, q):
+    os.environ["GLOO_SOCKET_IFNAME"] = os.environ.get("GLOO_SOCKET_IFNAME", "lo")
+    import datetime
+
+    import torch.distributed as dist
+
+    from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+    from sagemaker_xgboost_container_amd.models import trainer
+    from sagemaker_xgboost_container_amd.parallel.comm import Communicator
+
+    dist.init_process_group(
+        backend="gloo", init_method=f"tcp://127.0.0.1:{port}", rank=rank, world_size=world,
+        timeout=datetime.timedelta(seconds=120),
+    )
+    comm = Communicator()
+    rng = np.random.default_rng(7)
+    X = rng.normal(size=(1000, 6)).astype(np.float32)
+    y = (X[:, 0] - 0.3 * X[:, 2] > 0).astype(np.float32)
+    # deliberately UNEVEN shards: naive unweighted averaging would be wrong
+    cut = 700
+    sl = slice(0, cut) if rank == 0 else slice(cut, None)
+    res = {}
+    bst = trainer.train(
+        {"objective": "binary:logistic", "max_depth": 3, "eta": 0.3, "device": "cpu",
+         "eval_metric": ["logloss", "rmse", "auc"]},
+        DMatrix(X[sl], label=y[sl]),
+        num_boost_round=4,
+        evals=[(DMatrix(X[sl], label=y[sl]), "train")],
+        evals_result=res,
+        verbose_eval=False,
+        comm=comm,
+    )
+    out = {m: res["train"][m][-1] for m in ("logloss", "rmse", "auc")}
+    if rank == 0:
+        out["pred_full"] = bst.predict(X).tolist()
+    q.put((rank, out))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_distributed_eval_metric_aggregation():
+    """Eval metrics must be identical on every rank (early stopping depends
+    on it), and ratio-of-sums metrics must equal the full-data value exactly
+    despite uneven shards (700/300 rows)."""
+    import torch
+
+    from sagemaker_xgboost_container_amd.models import eval_metrics
+
+    port = _find_open_ports(1)[0]
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_eval_agg_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = dict(q.get(timeout=300) for _ in range(2))
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+
+    for m in ("logloss", "rmse", "auc"):
+        assert results[0][m] == pytest.approx(results[1][m], abs=1e-12), m
+
+    # cross-check ratio-of-sums metrics against a direct full-data computation
+    rng = np.random.default_rng(7)
+    X = rng.normal(size=(1000, 6)).astype(np.float32)
+    y = (X[:, 0] - 0.3 * X[:, 2] > 0).astype(np.float32)
+    prob = torch.tensor(np.asarray(results[0]["pred_full"]), dtype=torch.float64)
+    yt = torch.tensor(y, dtype=torch.float64)
+    full_logloss = eval_metrics.logloss(prob, yt)
+    assert results[0]["logloss"] == pytest.approx(full_logloss, abs=2e-4)
