@@ -597,7 +597,8 @@ __global__ __launch_bounds__(HIST_BLOCK) void partition_device_kernel(
     float2* __restrict__ dst_gh, int* __restrict__ dst_rows,
     const LevelNode* __restrict__ nodes, const int* __restrict__ part_prefix,
     const LevelWork* __restrict__ work, const float* __restrict__ split_packed,
-    int* __restrict__ counters, int k, int nfeat, int missing_bin) {
+    int* __restrict__ counters, int k, int nfeat, int missing_bin,
+    int copy_payload) {
   __shared__ int ldest[CPART_TILE];
   __shared__ int lcnt, rcnt, lbase, rbase;
   const int total = work->part_total;
@@ -633,7 +634,10 @@ __global__ __launch_bounds__(HIST_BLOCK) void partition_device_kernel(
         rbase = atomicAdd(&counters[slot * 2 + 1], rcnt);
       }
       __syncthreads();
-      if ((nfeat & 3) == 0 && sizeof(BinT) == 1) {
+      // the deepest level's partition only feeds leaf_update (row ids +
+      // counts): bins/gh copies are skipped there (copy_payload == 0),
+      // dropping ~90% of that level's partition traffic.
+      if (copy_payload && (nfeat & 3) == 0 && sizeof(BinT) == 1) {
         const int nd = nfeat >> 2;
         int log2p = 0;
         while ((1 << log2p) < nd) ++log2p;
@@ -649,7 +653,7 @@ __global__ __launch_bounds__(HIST_BLOCK) void partition_device_kernel(
               d >= 0 ? (long long)node.start + lbase + d : (long long)node.end - 1 - rbase - (~d);
           db4[dst * nd + f4] = sb4[(tile + i) * (long long)nd + f4];
         }
-      } else {
+      } else if (copy_payload) {
         int log2p = 0;
         while ((1 << log2p) < nfeat) ++log2p;
         const int mask = (1 << log2p) - 1;
@@ -663,14 +667,16 @@ __global__ __launch_bounds__(HIST_BLOCK) void partition_device_kernel(
           dst_bins[dst * (long long)nfeat + f] = src_bins[(tile + i) * (long long)nfeat + f];
         }
       }
-      for (int u = threadIdx.x; u < tile_n * 2; u += blockDim.x) {
-        const int i = u >> 1;
-        const int half = u & 1;
-        const int d = ldest[i];
-        const long long dst =
-            d >= 0 ? (long long)node.start + lbase + d : (long long)node.end - 1 - rbase - (~d);
-        reinterpret_cast<float*>(dst_gh)[dst * 2 + half] =
-            reinterpret_cast<const float*>(src_gh)[(tile + i) * 2 + half];
+      if (copy_payload) {
+        for (int u = threadIdx.x; u < tile_n * 2; u += blockDim.x) {
+          const int i = u >> 1;
+          const int half = u & 1;
+          const int d = ldest[i];
+          const long long dst =
+              d >= 0 ? (long long)node.start + lbase + d : (long long)node.end - 1 - rbase - (~d);
+          reinterpret_cast<float*>(dst_gh)[dst * 2 + half] =
+              reinterpret_cast<const float*>(src_gh)[(tile + i) * 2 + half];
+        }
       }
       for (int i = threadIdx.x; i < tile_n; i += blockDim.x) {
         const int d = ldest[i];
@@ -1207,7 +1213,8 @@ void grow_partition_level(torch::Tensor src_bins, torch::Tensor src_gh, torch::T
                           torch::Tensor dst_bins, torch::Tensor dst_gh, torch::Tensor dst_rows,
                           torch::Tensor nodes, torch::Tensor part_prefix, torch::Tensor work,
                           torch::Tensor split_packed, torch::Tensor counters, int64_t k,
-                          int64_t nfeat, int64_t missing_bin, int64_t grid) {
+                          int64_t nfeat, int64_t missing_bin, int64_t grid,
+                          int64_t copy_payload) {
   auto stream = current_stream();
   if (src_bins.scalar_type() == torch::kUInt8) {
     hipLaunchKernelGGL(partition_device_kernel<unsigned char>, dim3((int)grid), dim3(HIST_BLOCK),
@@ -1217,7 +1224,7 @@ void grow_partition_level(torch::Tensor src_bins, torch::Tensor src_gh, torch::T
                        dst_rows.data_ptr<int>(), (const LevelNode*)nodes.data_ptr<int>(),
                        part_prefix.data_ptr<int>(), (const LevelWork*)work.data_ptr<int>(),
                        split_packed.data_ptr<float>(), counters.data_ptr<int>(), (int)k,
-                       (int)nfeat, (int)missing_bin);
+                       (int)nfeat, (int)missing_bin, (int)copy_payload);
   } else {
     hipLaunchKernelGGL(partition_device_kernel<short>, dim3((int)grid), dim3(HIST_BLOCK), 0,
                        stream, src_bins.data_ptr<short>(), (const float2*)src_gh.data_ptr<float>(),
@@ -1225,7 +1232,8 @@ void grow_partition_level(torch::Tensor src_bins, torch::Tensor src_gh, torch::T
                        (float2*)dst_gh.data_ptr<float>(), dst_rows.data_ptr<int>(),
                        (const LevelNode*)nodes.data_ptr<int>(), part_prefix.data_ptr<int>(),
                        (const LevelWork*)work.data_ptr<int>(), split_packed.data_ptr<float>(),
-                       counters.data_ptr<int>(), (int)k, (int)nfeat, (int)missing_bin);
+                       counters.data_ptr<int>(), (int)k, (int)nfeat, (int)missing_bin,
+                       (int)copy_payload);
   }
 }
 
@@ -1330,16 +1338,18 @@ void grow_tree_enqueue(
     hipLaunchKernelGGL(split_reduce_kernel, dim3(k), dim3(SPLIT_BLOCK), 0, stream,
                        (const SplitCand*)cands.data_ptr<float>(), splits_d, k, (int)nfeat);
 
+    const int copy_payload = d < (int)D - 1 ? 1 : 0;  // last level: rows+counts only
     if (u8) {
       hipLaunchKernelGGL(partition_device_kernel<unsigned char>, dim3((int)part_grid),
                          dim3(HIST_BLOCK), 0, stream, (const unsigned char*)src_bins, src_gh,
                          src_rows, (unsigned char*)dst_bins, dst_gh, dst_rows, nodes_d, pp_d,
-                         work_d, splits_d, counts_d, k, (int)nfeat, (int)missing_bin);
+                         work_d, splits_d, counts_d, k, (int)nfeat, (int)missing_bin,
+                         copy_payload);
     } else {
       hipLaunchKernelGGL(partition_device_kernel<short>, dim3((int)part_grid), dim3(HIST_BLOCK),
                          0, stream, (const short*)src_bins, src_gh, src_rows, (short*)dst_bins,
                          dst_gh, dst_rows, nodes_d, pp_d, work_d, splits_d, counts_d, k,
-                         (int)nfeat, (int)missing_bin);
+                         (int)nfeat, (int)missing_bin, copy_payload);
     }
   }
 }

@@ -605,6 +605,7 @@ class DeviceGrower:
                 src_bins, src_gh, src_rows, st._bins[dst], st._gh[dst], st._rows[dst],
                 nodes_d, self.pp[d], self.work[d], splits_d, counts_d,
                 k, f, missing_bin, _GROW_PART_GRID,
+                1 if d < self.D - 1 else 0,  # last level feeds leaf_update only
             )
         # (the grower decides _level0 from whether the root split)
 
