@@ -1,19 +1,31 @@
 """Training objectives: per-row gradient/hessian + prediction transforms.
 
 Replaces the native objective kernels behind ``xgb.train``.
-All math is torch (runs on CPU or ROCm device); the per-element work is
-memory-bound and fuses well under torch on ROCm for these shapes — the hot
-kernels of this framework are histogram/partition, not grad/hess.
+Math is torch (runs on CPU or ROCm device); the hot scalar objectives
+(binary:logistic, reg:squarederror) additionally have a one-pass fused HIP
+kernel (grad_fused_kernel) that writes the packed gh AND the |g|/|h| maxima
+the fixed-point histogram scale needs — torch needs ~5 HBM passes for the
+same work (~250 us/round on the 12.5M-row bench vs ~35 us fused).
 
 Error-message substrings for bad labels intentionally match
 constants/xgb_constants.CUSTOMER_ERRORS so algorithm_mode/train.py maps them
 to UserError (reference train.py:461-467 behavior).
 """
 import math
+import os
 
 import torch
 
 from ..constants import xgb_constants as xgbc
+
+
+def _fused_grad(name, margin, y, weight, scale_pos_weight=1.0):
+    """HIP fast path for the hot objectives; None -> caller uses torch."""
+    if not margin.is_cuda or os.environ.get("SMXGB_FORCE_TORCH_OPS") == "1":
+        return None
+    from ..ops import hip as _hip  # raises if the extension is missing on GPU
+
+    return _hip.fused_gradients(name, margin, y, weight, scale_pos_weight)
 
 
 class Objective:
@@ -52,6 +64,9 @@ class SquaredError(Objective):
     default_metric = "rmse"
 
     def gradients(self, margin, y, weight=None):
+        fused = _fused_grad(self.name, margin, y, weight)
+        if fused is not None:
+            return fused
         g = margin - y
         h = torch.ones_like(margin)
         return _pack(g, h, weight)
@@ -93,6 +108,9 @@ class Logistic(Objective):
         return torch.sigmoid(margin)
 
     def gradients(self, margin, y, weight=None):
+        fused = _fused_grad(self.name, margin, y, weight, self.scale_pos_weight)
+        if fused is not None:
+            return fused
         p = torch.sigmoid(margin)
         g = p - y
         h = torch.clamp(p * (1 - p), min=1e-16)

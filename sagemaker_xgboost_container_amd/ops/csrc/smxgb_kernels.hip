@@ -351,6 +351,59 @@ __global__ __launch_bounds__(HIST_BLOCK) void partition_compact_kernel(
   }
 }
 
+// fused per-round gradient computation: one HBM pass writes the packed
+// (g, h) pairs AND per-block |g|/|h| maxima (for the fixed-point histogram
+// scale), replacing ~5 torch elementwise/reduce passes (~250 us/round on the
+// 12.5M-row bench; this kernel is ~35 us).
+// mode 0: binary:logistic / reg:logistic — g = sigmoid(m) - y,
+//         h = max(p(1-p), 1e-16), scale_pos_weight applied to y == 1
+// mode 1: reg:squarederror — g = m - y, h = 1
+__global__ __launch_bounds__(HIST_BLOCK) void grad_fused_kernel(
+    const float* __restrict__ margin, const float* __restrict__ y,
+    const float* __restrict__ w, float2* __restrict__ gh,
+    float2* __restrict__ pmax, long long n, int mode, float spw) {
+  __shared__ float red[HIST_BLOCK * 2];
+  float gmax = 0.f, hmax = 0.f;
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
+    const float m = margin[i];
+    const float yi = y[i];
+    float g, h;
+    if (mode == 0) {
+      const float p = 1.0f / (1.0f + expf(-m));
+      g = p - yi;
+      h = fmaxf(p * (1.0f - p), 1e-16f);
+      if (spw != 1.0f && yi == 1.0f) {
+        g *= spw;
+        h *= spw;
+      }
+    } else {
+      g = m - yi;
+      h = 1.0f;
+    }
+    if (w) {
+      const float wi = w[i];
+      g *= wi;
+      h *= wi;
+    }
+    gh[i] = make_float2(g, h);
+    gmax = fmaxf(gmax, fabsf(g));
+    hmax = fmaxf(hmax, fabsf(h));
+  }
+  red[threadIdx.x] = gmax;
+  red[threadIdx.x + HIST_BLOCK] = hmax;
+  __syncthreads();
+  for (int off = HIST_BLOCK / 2; off > 0; off >>= 1) {
+    if (threadIdx.x < off) {
+      red[threadIdx.x] = fmaxf(red[threadIdx.x], red[threadIdx.x + off]);
+      red[threadIdx.x + HIST_BLOCK] =
+          fmaxf(red[threadIdx.x + HIST_BLOCK], red[threadIdx.x + HIST_BLOCK + off]);
+    }
+    __syncthreads();
+  }
+  if (threadIdx.x == 0) pmax[blockIdx.x] = make_float2(red[0], red[HIST_BLOCK]);
+}
+
 // leaf scatter from compact row-id buffers
 __global__ __launch_bounds__(HIST_BLOCK) void leaf_update_compact_kernel(
     const int* __restrict__ rows0, const int* __restrict__ rows1,
@@ -1354,6 +1407,17 @@ void grow_tree_enqueue(
   }
 }
 
+void grad_fused(torch::Tensor margin, torch::Tensor y, torch::Tensor w, torch::Tensor gh,
+                torch::Tensor pmax, int64_t mode, double spw) {
+  CHECK_GPU(margin);
+  const long long n = margin.numel();
+  const int grid = (int)pmax.size(0);
+  hipLaunchKernelGGL(grad_fused_kernel, dim3(grid), dim3(HIST_BLOCK), 0, current_stream(),
+                     margin.data_ptr<float>(), y.data_ptr<float>(),
+                     w.numel() ? w.data_ptr<float>() : nullptr, (float2*)gh.data_ptr<float>(),
+                     (float2*)pmax.data_ptr<float>(), n, (int)mode, (float)spw);
+}
+
 void find_splits(torch::Tensor hist, torch::Tensor parent, torch::Tensor nbins,
                  torch::Tensor feat_mask, torch::Tensor monotone, torch::Tensor cands,
                  torch::Tensor out, int64_t k, int64_t f, int64_t stride, int64_t has_missing,
@@ -1397,5 +1461,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("leaf_update_compact", &leaf_update_compact, "leaf scatter from compact row ids");
   m.def("leaf_update", &leaf_update, "batched leaf value scatter into margins");
   m.def("predict_forest", &predict_forest, "batched dense forest traversal");
+  m.def("grad_fused", &grad_fused, "one-pass gradient + absmax partials");
   m.attr("_built_for") = "gfx950";
 }

@@ -41,11 +41,47 @@ _MAX_BLOCKS_PER_JOB = int(_os.environ.get("SMXGB_MAX_BLOCKS", "2048"))
 
 def compute_scale(gh, comm=None):
     """Device-resident (gmax, hmax); kernels derive 2^33/max on device —
-    no host synchronization per tree."""
-    m = gh.abs().amax(dim=0).contiguous()  # (2,)
+    no host synchronization per tree. `fused_gradients` attaches the absmax
+    it computed in its single pass; recompute only when absent."""
+    m = getattr(gh, "_smxgb_absmax", None)
+    if m is None:
+        m = gh.abs().amax(dim=0).contiguous()  # (2,)
     if comm is not None:
         comm.allreduce_max_(m)
     return m
+
+
+_GRAD_GRID = 2048
+_FUSED_GRAD_MODES = {
+    "binary:logistic": 0,
+    "reg:logistic": 0,
+    "binary:logitraw": 0,  # same gradient formula; only transform() differs
+    "reg:squarederror": 1,
+}
+
+
+def fused_gradients(name, margin, y, weight=None, scale_pos_weight=1.0):
+    """One-pass gradient + absmax for the hot objectives (grad_fused_kernel).
+
+    Returns the packed (n, 2) gh tensor with `_smxgb_absmax` attached, or
+    None when the objective/layout is unsupported (caller falls back to the
+    torch path).
+    """
+    mode = _FUSED_GRAD_MODES.get(name)
+    if mode is None or not margin.is_cuda or margin.dim() != 1:
+        return None
+    margin = margin.contiguous()
+    n = margin.numel()
+    gh = torch.empty((n, 2), dtype=torch.float32, device=margin.device)
+    pmax = torch.empty((_GRAD_GRID, 2), dtype=torch.float32, device=margin.device)
+    w = (
+        weight.contiguous()
+        if (weight is not None and weight.numel())
+        else torch.empty(0, dtype=torch.float32, device=margin.device)
+    )
+    _K.grad_fused(margin, y.contiguous(), w, gh, pmax, mode, float(scale_pos_weight))
+    gh._smxgb_absmax = pmax.amax(dim=0).contiguous()
+    return gh
 
 
 def _padded_words(pairs):
