@@ -353,7 +353,11 @@ class TreeState:
         n = qm.num_row
         if sample_rows is None:
             cap = n
-            self._rows_init = torch.arange(n, dtype=torch.int32, device=device)
+            rid = getattr(qm, "_arange_cache", None)
+            if rid is None or rid.numel() != n:
+                rid = torch.arange(n, dtype=torch.int32, device=device)
+                qm._arange_cache = rid
+            self._rows_init = rid
             self._bins_init = qm.bins
             self._gh_init = gh.contiguous()
         else:
