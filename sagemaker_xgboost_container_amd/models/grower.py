@@ -201,10 +201,15 @@ class HistGrower:
             return tree, leaf_jobs
 
         tree = Tree()
+        cached = getattr(gh, "_smxgb_rootsum", None) if rows is None else None
         root_sum = (
-            gh.to(torch.float64).sum(0)
-            if rows is None
-            else gh.index_select(0, rows.long()).to(torch.float64).sum(0)
+            cached.clone()
+            if cached is not None
+            else (
+                gh.to(torch.float64).sum(0)
+                if rows is None
+                else gh.index_select(0, rows.long()).to(torch.float64).sum(0)
+            )
         )
         root_sum = self._allreduce(root_sum)
         G, H = float(root_sum[0]), float(root_sum[1])

@@ -361,9 +361,12 @@ __global__ __launch_bounds__(HIST_BLOCK) void partition_compact_kernel(
 __global__ __launch_bounds__(HIST_BLOCK) void grad_fused_kernel(
     const float* __restrict__ margin, const float* __restrict__ y,
     const float* __restrict__ w, float2* __restrict__ gh,
-    float2* __restrict__ pmax, long long n, int mode, float spw) {
+    float2* __restrict__ pmax, double2* __restrict__ psum, long long n, int mode,
+    float spw) {
   __shared__ float red[HIST_BLOCK * 2];
+  __shared__ double dred[HIST_BLOCK * 2];
   float gmax = 0.f, hmax = 0.f;
+  double gsum = 0.0, hsum = 0.0;
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
     const float m = margin[i];
@@ -389,19 +392,28 @@ __global__ __launch_bounds__(HIST_BLOCK) void grad_fused_kernel(
     gh[i] = make_float2(g, h);
     gmax = fmaxf(gmax, fabsf(g));
     hmax = fmaxf(hmax, fabsf(h));
+    gsum += (double)g;
+    hsum += (double)h;
   }
   red[threadIdx.x] = gmax;
   red[threadIdx.x + HIST_BLOCK] = hmax;
+  dred[threadIdx.x] = gsum;
+  dred[threadIdx.x + HIST_BLOCK] = hsum;
   __syncthreads();
   for (int off = HIST_BLOCK / 2; off > 0; off >>= 1) {
     if (threadIdx.x < off) {
       red[threadIdx.x] = fmaxf(red[threadIdx.x], red[threadIdx.x + off]);
       red[threadIdx.x + HIST_BLOCK] =
           fmaxf(red[threadIdx.x + HIST_BLOCK], red[threadIdx.x + HIST_BLOCK + off]);
+      dred[threadIdx.x] += dred[threadIdx.x + off];
+      dred[threadIdx.x + HIST_BLOCK] += dred[threadIdx.x + HIST_BLOCK + off];
     }
     __syncthreads();
   }
-  if (threadIdx.x == 0) pmax[blockIdx.x] = make_float2(red[0], red[HIST_BLOCK]);
+  if (threadIdx.x == 0) {
+    pmax[blockIdx.x] = make_float2(red[0], red[HIST_BLOCK]);
+    psum[blockIdx.x] = make_double2(dred[0], dred[HIST_BLOCK]);
+  }
 }
 
 // leaf scatter from compact row-id buffers
@@ -1408,14 +1420,15 @@ void grow_tree_enqueue(
 }
 
 void grad_fused(torch::Tensor margin, torch::Tensor y, torch::Tensor w, torch::Tensor gh,
-                torch::Tensor pmax, int64_t mode, double spw) {
+                torch::Tensor pmax, torch::Tensor psum, int64_t mode, double spw) {
   CHECK_GPU(margin);
   const long long n = margin.numel();
   const int grid = (int)pmax.size(0);
   hipLaunchKernelGGL(grad_fused_kernel, dim3(grid), dim3(HIST_BLOCK), 0, current_stream(),
                      margin.data_ptr<float>(), y.data_ptr<float>(),
                      w.numel() ? w.data_ptr<float>() : nullptr, (float2*)gh.data_ptr<float>(),
-                     (float2*)pmax.data_ptr<float>(), n, (int)mode, (float)spw);
+                     (float2*)pmax.data_ptr<float>(), (double2*)psum.data_ptr<double>(), n,
+                     (int)mode, (float)spw);
 }
 
 void find_splits(torch::Tensor hist, torch::Tensor parent, torch::Tensor nbins,

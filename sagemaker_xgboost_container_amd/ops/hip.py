@@ -74,13 +74,17 @@ def fused_gradients(name, margin, y, weight=None, scale_pos_weight=1.0):
     n = margin.numel()
     gh = torch.empty((n, 2), dtype=torch.float32, device=margin.device)
     pmax = torch.empty((_GRAD_GRID, 2), dtype=torch.float32, device=margin.device)
+    psum = torch.empty((_GRAD_GRID, 2), dtype=torch.float64, device=margin.device)
     w = (
         weight.contiguous()
         if (weight is not None and weight.numel())
         else torch.empty(0, dtype=torch.float32, device=margin.device)
     )
-    _K.grad_fused(margin, y.contiguous(), w, gh, pmax, mode, float(scale_pos_weight))
+    _K.grad_fused(margin, y.contiguous(), w, gh, pmax, psum, mode, float(scale_pos_weight))
     gh._smxgb_absmax = pmax.amax(dim=0).contiguous()
+    # deterministic full-data (G, H): fixed per-block partial order + one
+    # torch sum over the 2048 partials
+    gh._smxgb_rootsum = psum.sum(dim=0)
     return gh
 
 
@@ -569,7 +573,10 @@ class DeviceGrower:
         reg_lambda, reg_alpha, gamma, mcw = split_params
 
         self.counts.zero_()
-        root_gh = st._gh_init.to(torch.float64).sum(0)
+        # full-data (G, H): the fused gradient kernel attaches its one-pass
+        # partial-sum result; recompute only for subsampled/torch-path gh
+        root_gh = getattr(st._gh_init, "_smxgb_rootsum", None)
+        root_gh = root_gh.clone() if root_gh is not None else st._gh_init.to(torch.float64).sum(0)
         if comm is not None:
             comm.allreduce_(root_gh)
         self.node_gh[0] = root_gh.to(torch.float32)
