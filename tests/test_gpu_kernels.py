@@ -338,9 +338,12 @@ class TestFusedGradients:
         # attached absmax must equal the true column maxima
         expect = gh_ref.abs().amax(dim=0)
         torch.testing.assert_close(gh_fused._smxgb_absmax.cpu(), expect, rtol=2e-5, atol=2e-6)
-        # attached root (G, H) must equal the f64 full-data sums
-        expect_sum = gh_ref.to(torch.float64).sum(0)
-        torch.testing.assert_close(gh_fused._smxgb_rootsum.cpu(), expect_sum, rtol=1e-6, atol=1e-4)
+        # attached root (G, H) must equal the f64 sum of the gh the kernel
+        # itself produced (torch's sigmoid differs from expf by ~1e-8 per
+        # element, which accumulates to ~1e-2 over 1M rows — comparing sums
+        # across the two gradient implementations would test nothing useful)
+        expect_sum = gh_fused.to(torch.float64).sum(0)
+        torch.testing.assert_close(gh_fused._smxgb_rootsum, expect_sum, rtol=1e-10, atol=1e-8)
 
     def test_unsupported_objective_falls_back(self):
         from sagemaker_xgboost_container_amd.ops import hip as hip_ops
