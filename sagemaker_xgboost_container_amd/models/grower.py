@@ -223,16 +223,23 @@ class HistGrower:
         tree.finalize()
         return tree, leaf_jobs
 
+    def _device_grower_for(self, slot):
+        """One DeviceGrower (own heap buffers) per (depth, matrix, slot)."""
+        key = (self.p.max_depth, id(self.qm), slot)
+        growers = getattr(self, "_device_growers", None)
+        if growers is None:
+            growers = {}
+            self._device_growers = growers
+        dg = growers.get(key)
+        if dg is None:
+            dg = self.backend.DeviceGrower(self.state, self.p.max_depth)
+            growers[key] = dg
+        return dg
+
     def _grow_depthwise_device(self, gh, scale, cap, tree_mask):
         """Consume the DeviceGrower's one-readback result into a Tree."""
         p = self.p
-        qm = self.qm
-        D = p.max_depth
-        key = (D, id(self.state.qm))
-        if getattr(self, "_device_grower_key", None) != key:
-            self._device_grower = self.backend.DeviceGrower(self.state, D)
-            self._device_grower_key = key
-        dg = self._device_grower
+        dg = self._device_grower_for(0)
         dg.state = self.state  # fresh per-tree compact state
         if tree_mask is not None:
             dg.mask = tree_mask.to(torch.uint8).contiguous()
@@ -240,7 +247,71 @@ class HistGrower:
             scale, (p.reg_lambda, p.reg_alpha, p.gamma, p.min_child_weight), self.comm
         )
         self.state._level0 = not bool(splits_np[0, 0] > 0)
+        return self._build_tree_from_heap(splits_np, counts_np, root_np, cap)
 
+    def device_async_ok(self):
+        """True when a round's independent trees (multiclass / bagging) can
+        be enqueued back-to-back via grow_async/grow_finish."""
+        import os as _os
+
+        p = self.p
+        return (
+            self.comm is None
+            and hasattr(self.backend, "DeviceGrower")
+            and _os.environ.get("SMXGB_NO_DEVICE_GROW") != "1"
+            and _os.environ.get("SMXGB_PIPELINE") != "v1"
+            and p.grow_policy == "depthwise"
+            and p.max_leaves == 0
+            and 1 <= p.max_depth <= 10
+            and self.monotone is None
+            and self.inter_sets is None
+            and p.colsample_bylevel >= 1.0
+            and p.colsample_bynode >= 1.0
+        )
+
+    def grow_async(self, gh, slot=0):
+        """Enqueue one independent tree of a round without waiting.
+
+        Requires device_async_ok(). Each `slot` uses its own compact buffers
+        and heap, so several trees can be in flight on the stream at once —
+        the GPU never idles while the host builds the previous class's tree.
+        Returns an opaque handle for grow_finish().
+        """
+        p = self.p
+        qm = self.qm
+        n = qm.num_row
+        if p.subsample < 1.0:
+            keep = torch.rand(n, device=self.device, generator=self.generator) < p.subsample
+            rows = keep.nonzero(as_tuple=True)[0].to(torch.int32)
+        else:
+            rows = None
+        state = self.backend.make_tree_state(qm, gh, rows, slot=slot)
+        self.state = state
+        scale = self.backend.compute_scale(gh, comm=None)
+        tree_mask = self._sample_features(p.colsample_bytree, None)
+        dg = self._device_grower_for(slot)
+        dg.state = state
+        if tree_mask is not None:
+            dg.mask = tree_mask.to(torch.uint8).contiguous()
+        ev = dg.grow_enqueue(scale, (p.reg_lambda, p.reg_alpha, p.gamma, p.min_child_weight))
+        return {"dg": dg, "ev": ev, "state": state, "cap": state.cap}
+
+    def grow_finish(self, handle):
+        """Wait for a grow_async tree and build its host Tree."""
+        splits_np, counts_np, root_np = handle["dg"].grow_wait(handle["ev"])
+        state = handle["state"]
+        state._level0 = not bool(splits_np[0, 0] > 0)
+        self.state = state  # so the caller's update_margins hits this slot
+        tree, leaf_jobs = self._build_tree_from_heap(
+            splits_np, counts_np, root_np, handle["cap"]
+        )
+        tree.finalize()
+        return tree, leaf_jobs
+
+    def _build_tree_from_heap(self, splits_np, counts_np, root_np, cap):
+        p = self.p
+        qm = self.qm
+        D = p.max_depth
         if not hasattr(qm, "_cuts_np"):
             qm._cuts_np = qm.cuts.cpu().numpy()
             qm._cut_ptr_np = qm.cut_ptr.cpu().numpy()

@@ -303,7 +303,31 @@ def train(
 
             round_trees = []
             round_info = []
+            # A round's trees (one per class) all derive from the same margin
+            # state, so they are independent: on the device path they are
+            # enqueued back-to-back (per-class buffer slots) and drained in
+            # order — the GPU never idles while the host builds tree objects.
+            pipeline_classes = (
+                n_outputs > 1
+                and dart is None
+                and hasattr(grower, "device_async_ok")
+                and grower.device_async_ok()
+            )
             for _parallel in range(num_parallel_tree):
+                if pipeline_classes:
+                    handles = [
+                        grower.grow_async(gh[:, cls, :].contiguous(), slot=cls)
+                        for cls in range(n_outputs)
+                    ]
+                    for cls, handle in enumerate(handles):
+                        tree, leaf_jobs = grower.grow_finish(handle)
+                        grower.state.update_margins(margin[:, cls], leaf_jobs)
+                        round_trees.append(tree)
+                        round_info.append(cls)
+                        for es in eval_sets:
+                            if not es.is_train:
+                                es.margin[:, cls] += backend.predict_tree(tree, es.X)
+                    continue
                 for cls in range(n_outputs):
                     gh_cls = gh if n_outputs == 1 else gh[:, cls, :].contiguous()
                     tree, leaf_jobs = grower.grow(gh_cls)

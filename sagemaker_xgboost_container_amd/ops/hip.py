@@ -351,7 +351,7 @@ class TreeState:
     the tree is row-subsampled.
     """
 
-    def __init__(self, qm, gh, sample_rows=None):
+    def __init__(self, qm, gh, sample_rows=None, slot=0):
         self.qm = qm
         device = qm.bins.device
         n = qm.num_row
@@ -371,7 +371,13 @@ class TreeState:
             self._bins_init = qm.bins.index_select(0, idx).contiguous()
             self._gh_init = gh.index_select(0, idx).contiguous()
         self.cap = cap
-        cache = getattr(qm, "_compact_cache", None)
+        # `slot` keys independent buffer sets so several trees of one round
+        # (multiclass / bagging) can be in flight on the GPU at once
+        caches = getattr(qm, "_compact_cache", None)
+        if caches is None:
+            caches = {}
+            qm._compact_cache = caches
+        cache = caches.get(slot)
         if cache is None or cache[0].shape[0] < cap:
             cache = (
                 torch.empty((cap, qm.num_col), dtype=qm.bins.dtype, device=device),
@@ -381,7 +387,7 @@ class TreeState:
                 torch.empty(cap, dtype=torch.int32, device=device),
                 torch.empty(cap, dtype=torch.int32, device=device),
             )
-            qm._compact_cache = cache
+            caches[slot] = cache
         self._bins = [cache[0], cache[1]]
         self._gh = [cache[2], cache[3]]
         self._rows = [cache[4], cache[5]]
@@ -562,9 +568,11 @@ class DeviceGrower:
         if not hasattr(qm, "_nbins_i32"):
             qm._nbins_i32 = qm.nbins.to(torch.int32).contiguous()
 
-    def grow(self, scale, split_params, comm=None):
-        """Enqueue the full tree; returns (splits_np [H,6], counts_np [H,2],
-        root_gh_np [2])."""
+    def grow_enqueue(self, scale, split_params):
+        """Enqueue the full tree plus the non-blocking heap readback into
+        pinned host buffers; returns a torch.cuda.Event to wait on. Lets the
+        caller enqueue several independent trees (multiclass / bagging
+        rounds) back-to-back so the GPU never idles between them."""
         st = self.state
         qm = self.qm
         f = qm.num_col
@@ -577,28 +585,63 @@ class DeviceGrower:
         # partial-sum result; recompute only for subsampled/torch-path gh
         root_gh = getattr(st._gh_init, "_smxgb_rootsum", None)
         root_gh = root_gh.clone() if root_gh is not None else st._gh_init.to(torch.float64).sum(0)
-        if comm is not None:
-            comm.allreduce_(root_gh)
         self.node_gh[0] = root_gh.to(torch.float32)
 
-        if comm is None:
-            # whole tree enqueued from ONE extension call
-            _K.grow_tree_enqueue(
-                st._bins_init, st._gh_init, st._rows_init,
-                st._bins[0], st._gh[0], st._rows[0],
-                st._bins[1], st._gh[1], st._rows[1],
-                self.nodes, self.node_gh, self.splits, self.counts,
-                self.hist_f32, self.acc, self.cands, self.hp, self.pp, self.work,
-                qm._nbins_i32, self.mask, scale,
-                self.D, st.cap, f, stride, self.n_groups, self.feats_per_group,
-                self.lds_words, 1 if qm.has_missing else 0, missing_bin,
-                _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB, _GROW_HIST_GRID, _GROW_PART_GRID,
-                reg_lambda, reg_alpha, gamma, mcw,
+        # whole tree enqueued from ONE extension call
+        _K.grow_tree_enqueue(
+            st._bins_init, st._gh_init, st._rows_init,
+            st._bins[0], st._gh[0], st._rows[0],
+            st._bins[1], st._gh[1], st._rows[1],
+            self.nodes, self.node_gh, self.splits, self.counts,
+            self.hist_f32, self.acc, self.cands, self.hp, self.pp, self.work,
+            qm._nbins_i32, self.mask, scale,
+            self.D, st.cap, f, stride, self.n_groups, self.feats_per_group,
+            self.lds_words, 1 if qm.has_missing else 0, missing_bin,
+            _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB, _GROW_HIST_GRID, _GROW_PART_GRID,
+            reg_lambda, reg_alpha, gamma, mcw,
+        )
+        if not hasattr(self, "_pinned"):
+            self._pinned = (
+                torch.empty_like(self.splits, device="cpu", pin_memory=True),
+                torch.empty_like(self.counts, device="cpu", pin_memory=True),
+                torch.empty((2,), dtype=torch.float32, device="cpu", pin_memory=True),
             )
-            splits_np = self.splits.cpu().numpy()
-            counts_np = self.counts.cpu().numpy()
-            root_np = self.node_gh[0].cpu().numpy()
-            return splits_np, counts_np, root_np
+        self._pinned[0].copy_(self.splits, non_blocking=True)
+        self._pinned[1].copy_(self.counts, non_blocking=True)
+        self._pinned[2].copy_(self.node_gh[0], non_blocking=True)
+        ev = torch.cuda.Event()
+        ev.record()
+        return ev
+
+    def grow_wait(self, ev):
+        """Wait for a grow_enqueue readback; returns (splits, counts, root)
+        numpy views of this grower's pinned buffers (valid until the next
+        grow_enqueue on the same instance)."""
+        ev.synchronize()
+        return (
+            self._pinned[0].numpy(),
+            self._pinned[1].numpy(),
+            self._pinned[2].numpy(),
+        )
+
+    def grow(self, scale, split_params, comm=None):
+        """Enqueue the full tree; returns (splits_np [H,6], counts_np [H,2],
+        root_gh_np [2])."""
+        if comm is None:
+            return self.grow_wait(self.grow_enqueue(scale, split_params))
+
+        st = self.state
+        qm = self.qm
+        f = qm.num_col
+        stride = qm.stride
+        missing_bin = stride - 1 if qm.has_missing else -1
+        reg_lambda, reg_alpha, gamma, mcw = split_params
+
+        self.counts.zero_()
+        root_gh = getattr(st._gh_init, "_smxgb_rootsum", None)
+        root_gh = root_gh.clone() if root_gh is not None else st._gh_init.to(torch.float64).sum(0)
+        comm.allreduce_(root_gh)
+        self.node_gh[0] = root_gh.to(torch.float32)
 
         for d in range(self.D):
             k = 1 << d
@@ -663,7 +706,7 @@ class DeviceGrower:
         return splits_np, counts_np, root_np
 
 
-def make_tree_state(qm, gh, sample_rows=None):
+def make_tree_state(qm, gh, sample_rows=None, slot=0):
     if _os.environ.get("SMXGB_PIPELINE") == "v1":
         return _V1TreeState(qm, gh, sample_rows)
-    return TreeState(qm, gh, sample_rows)
+    return TreeState(qm, gh, sample_rows, slot=slot)

@@ -360,5 +360,5 @@ class TreeState:
         update_margins(margin_col, self._bufs, leaf_jobs)
 
 
-def make_tree_state(qm, gh, sample_rows=None):
-    return TreeState(qm, gh, sample_rows)
+def make_tree_state(qm, gh, sample_rows=None, slot=0):
+    return TreeState(qm, gh, sample_rows)  # CPU reference: slot unused
