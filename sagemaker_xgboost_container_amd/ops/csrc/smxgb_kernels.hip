@@ -562,9 +562,12 @@ __device__ inline int find_slot(const int* __restrict__ prefix, int k, int vb) {
   return lo;
 }
 
-// hist over compact buffers driven by a device job table
-template <typename BinT>
-__global__ __launch_bounds__(HIST_BLOCK) void hist_device_kernel(
+// hist over compact buffers driven by a device job table.
+// BLOCK=256 pairs with grouped 56 KB slabs (2 blocks/CU); BLOCK=512 runs a
+// SINGLE full-feature slab (up to ~158 KB, 1 block/CU — the same 8 waves,
+// but bins/gh stream ONCE per row instead of once per feature group).
+template <typename BinT, int BLOCK>
+__global__ __launch_bounds__(BLOCK) void hist_device_kernel(
     const BinT* __restrict__ bins_c, const float2* __restrict__ gh_c,
     const LevelNode* __restrict__ nodes, const int* __restrict__ hist_prefix,
     const LevelWork* __restrict__ work, unsigned long long* __restrict__ out,
@@ -1227,30 +1230,47 @@ void grow_make_level(torch::Tensor cur, torch::Tensor splits, torch::Tensor coun
                      (int)max_blocks);
 }
 
+// dispatch helper: BLOCK is a compile-time launch bound (256 grouped-slab,
+// 512 single full-feature slab)
+template <typename BinT>
+static void launch_hist_device(int grid, int hist_block, size_t lds_bytes, hipStream_t stream,
+                               const BinT* bins_c, const float2* gh_c, const LevelNode* nodes,
+                               const int* hist_prefix, const LevelWork* work,
+                               unsigned long long* acc, int k, int nfeat, int stride,
+                               int n_groups, int feats_per_group, const float* gh_max,
+                               int rows_per_block) {
+  if (hist_block == 512) {
+    hipLaunchKernelGGL((hist_device_kernel<BinT, 512>), dim3(grid), dim3(512), lds_bytes, stream,
+                       bins_c, gh_c, nodes, hist_prefix, work, acc, k, nfeat, stride, n_groups,
+                       feats_per_group, gh_max, rows_per_block);
+  } else {
+    hipLaunchKernelGGL((hist_device_kernel<BinT, HIST_BLOCK>), dim3(grid), dim3(HIST_BLOCK),
+                       lds_bytes, stream, bins_c, gh_c, nodes, hist_prefix, work, acc, k, nfeat,
+                       stride, n_groups, feats_per_group, gh_max, rows_per_block);
+  }
+}
+
 void grow_hist_level(torch::Tensor bins_c, torch::Tensor gh_c, torch::Tensor nodes,
                      torch::Tensor hist_prefix, torch::Tensor work, torch::Tensor acc,
                      int64_t k, int64_t nfeat, int64_t stride, int64_t n_groups,
                      int64_t feats_per_group, torch::Tensor gh_max, int64_t rows_per_block,
-                     int64_t grid, int64_t lds_words) {
+                     int64_t grid, int64_t lds_words, int64_t hist_block) {
   const size_t lds_bytes = (size_t)lds_words * sizeof(unsigned long long);
   auto stream = current_stream();
   if (bins_c.scalar_type() == torch::kUInt8) {
-    hipLaunchKernelGGL(hist_device_kernel<unsigned char>, dim3((int)grid), dim3(HIST_BLOCK),
-                       lds_bytes, stream, bins_c.data_ptr<unsigned char>(),
-                       (const float2*)gh_c.data_ptr<float>(),
-                       (const LevelNode*)nodes.data_ptr<int>(), hist_prefix.data_ptr<int>(),
-                       (const LevelWork*)work.data_ptr<int>(),
-                       (unsigned long long*)acc.data_ptr<int64_t>(), (int)k, (int)nfeat,
-                       (int)stride, (int)n_groups, (int)feats_per_group,
-                       gh_max.data_ptr<float>(), (int)rows_per_block);
+    launch_hist_device<unsigned char>(
+        (int)grid, (int)hist_block, lds_bytes, stream, bins_c.data_ptr<unsigned char>(),
+        (const float2*)gh_c.data_ptr<float>(), (const LevelNode*)nodes.data_ptr<int>(),
+        hist_prefix.data_ptr<int>(), (const LevelWork*)work.data_ptr<int>(),
+        (unsigned long long*)acc.data_ptr<int64_t>(), (int)k, (int)nfeat, (int)stride,
+        (int)n_groups, (int)feats_per_group, gh_max.data_ptr<float>(), (int)rows_per_block);
   } else {
-    hipLaunchKernelGGL(hist_device_kernel<short>, dim3((int)grid), dim3(HIST_BLOCK), lds_bytes,
-                       stream, bins_c.data_ptr<short>(), (const float2*)gh_c.data_ptr<float>(),
-                       (const LevelNode*)nodes.data_ptr<int>(), hist_prefix.data_ptr<int>(),
-                       (const LevelWork*)work.data_ptr<int>(),
-                       (unsigned long long*)acc.data_ptr<int64_t>(), (int)k, (int)nfeat,
-                       (int)stride, (int)n_groups, (int)feats_per_group,
-                       gh_max.data_ptr<float>(), (int)rows_per_block);
+    launch_hist_device<short>(
+        (int)grid, (int)hist_block, lds_bytes, stream, bins_c.data_ptr<short>(),
+        (const float2*)gh_c.data_ptr<float>(), (const LevelNode*)nodes.data_ptr<int>(),
+        hist_prefix.data_ptr<int>(), (const LevelWork*)work.data_ptr<int>(),
+        (unsigned long long*)acc.data_ptr<int64_t>(), (int)k, (int)nfeat, (int)stride,
+        (int)n_groups, (int)feats_per_group, gh_max.data_ptr<float>(), (int)rows_per_block);
   }
 }
 
@@ -1316,7 +1336,8 @@ void grow_tree_enqueue(
     int64_t D, int64_t cap, int64_t nfeat, int64_t stride, int64_t n_groups,
     int64_t feats_per_group, int64_t lds_words, int64_t has_missing, int64_t missing_bin,
     int64_t rows_per_block, int64_t max_blocks, int64_t hist_grid, int64_t part_grid,
-    double reg_lambda, double reg_alpha, double gamma_, double min_child_weight) {
+    double reg_lambda, double reg_alpha, double gamma_, double min_child_weight,
+    int64_t hist_block) {
   CHECK_GPU(init_bins);
   auto stream = current_stream();
   const long long slots2 = (long long)nfeat * stride * 2;
@@ -1366,17 +1387,17 @@ void grow_tree_enqueue(
 
     hipMemsetAsync(acc.data_ptr<int64_t>(), 0, sizeof(int64_t) * (size_t)k * slots2, stream);
     if (u8) {
-      hipLaunchKernelGGL(hist_device_kernel<unsigned char>, dim3((int)hist_grid), dim3(HIST_BLOCK),
-                         lds_bytes, stream, (const unsigned char*)src_bins, src_gh, nodes_d, hp_d,
-                         work_d, (unsigned long long*)acc.data_ptr<int64_t>(), k, (int)nfeat,
-                         (int)stride, (int)n_groups, (int)feats_per_group,
-                         gh_max.data_ptr<float>(), (int)rows_per_block);
+      launch_hist_device<unsigned char>(
+          (int)hist_grid, (int)hist_block, lds_bytes, stream, (const unsigned char*)src_bins,
+          src_gh, nodes_d, hp_d, work_d, (unsigned long long*)acc.data_ptr<int64_t>(), k,
+          (int)nfeat, (int)stride, (int)n_groups, (int)feats_per_group,
+          gh_max.data_ptr<float>(), (int)rows_per_block);
     } else {
-      hipLaunchKernelGGL(hist_device_kernel<short>, dim3((int)hist_grid), dim3(HIST_BLOCK),
-                         lds_bytes, stream, (const short*)src_bins, src_gh, nodes_d, hp_d, work_d,
-                         (unsigned long long*)acc.data_ptr<int64_t>(), k, (int)nfeat, (int)stride,
-                         (int)n_groups, (int)feats_per_group, gh_max.data_ptr<float>(),
-                         (int)rows_per_block);
+      launch_hist_device<short>(
+          (int)hist_grid, (int)hist_block, lds_bytes, stream, (const short*)src_bins, src_gh,
+          nodes_d, hp_d, work_d, (unsigned long long*)acc.data_ptr<int64_t>(), k, (int)nfeat,
+          (int)stride, (int)n_groups, (int)feats_per_group, gh_max.data_ptr<float>(),
+          (int)rows_per_block);
     }
 
     {

@@ -104,6 +104,22 @@ def _feature_groups(nfeat, stride):
     return groups
 
 
+def _device_hist_plan(nfeat, stride):
+    """Pick the device-grower hist configuration: (groups, lds_words, block).
+
+    A single full-feature slab at 512 threads (1 block/CU = the same 8
+    waves as two 256-thread blocks on 56 KB slabs) streams bins/gh ONCE per
+    row instead of once per feature group; used whenever the padded slab
+    fits the 160 KB LDS. SMXGB_LDS_KB forces the grouped layout for sweeps.
+    """
+    if not _os.environ.get("SMXGB_LDS_KB"):
+        words = _padded_words(nfeat * stride)
+        if words * 8 <= 158 * 1024:
+            return [(0, nfeat)], words, 512
+    groups = _feature_groups(nfeat, stride)
+    return groups, _padded_words(max(fe - fs for fs, fe in groups) * stride), 256
+
+
 def _pack_jobs(fields_list):
     """fields_list: list of tuples of int32 words -> (jobs_dev, width)."""
     arr = np.asarray(fields_list, dtype=np.int32)
@@ -556,10 +572,9 @@ class DeviceGrower:
         self.hp = [torch.empty((1 << d) + 1, dtype=torch.int32, device=device) for d in range(max_depth)]
         self.pp = [torch.empty((1 << d) + 1, dtype=torch.int32, device=device) for d in range(max_depth)]
         self.work = torch.empty((max_depth, 2), dtype=torch.int32, device=device)
-        groups = _feature_groups(f, stride)
+        groups, self.lds_words, self.hist_block = _device_hist_plan(f, stride)
         self.n_groups = len(groups)
         self.feats_per_group = groups[0][1] - groups[0][0]
-        self.lds_words = _padded_words(max(fe - fs for fs, fe in groups) * stride)
         if feature_mask is None:
             self.mask = torch.empty(0, dtype=torch.uint8, device=device)
         else:
@@ -598,7 +613,7 @@ class DeviceGrower:
             self.D, st.cap, f, stride, self.n_groups, self.feats_per_group,
             self.lds_words, 1 if qm.has_missing else 0, missing_bin,
             _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB, _GROW_HIST_GRID, _GROW_PART_GRID,
-            reg_lambda, reg_alpha, gamma, mcw,
+            reg_lambda, reg_alpha, gamma, mcw, self.hist_block,
         )
         if not hasattr(self, "_pinned"):
             self._pinned = (
@@ -673,7 +688,7 @@ class DeviceGrower:
             _K.grow_hist_level(
                 src_bins, src_gh, nodes_d, self.hp[d], self.work[d], acc_d,
                 k, f, stride, self.n_groups, self.feats_per_group, scale,
-                _ROWS_PER_BLOCK, _GROW_HIST_GRID, self.lds_words,
+                _ROWS_PER_BLOCK, _GROW_HIST_GRID, self.lds_words, self.hist_block,
             )
             if comm is not None:
                 comm.allreduce_(acc_d)
