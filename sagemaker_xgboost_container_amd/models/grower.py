@@ -50,6 +50,13 @@ class GrowParams:
         self.colsample_bynode = float(p.get("colsample_bynode", 1.0))
         self.monotone_constraints = p.get("monotone_constraints")
         self.interaction_constraints = p.get("interaction_constraints")
+        # xgboost semantics: deterministic (bit-reproducible) histograms by
+        # default; "false"/"0" opts into the faster nondeterministic
+        # accumulation (single-process GPU only — distributed needs
+        # rank-consistent sums and always stays deterministic)
+        self.deterministic_histogram = str(
+            p.get("deterministic_histogram", "true")
+        ).lower() not in ("false", "0")
         if self.max_depth == 0 and self.max_leaves == 0 and self.grow_policy == "depthwise":
             self.max_depth = 6
 
@@ -225,14 +232,15 @@ class HistGrower:
 
     def _device_grower_for(self, slot):
         """One DeviceGrower (own heap buffers) per (depth, matrix, slot)."""
-        key = (self.p.max_depth, id(self.qm), slot)
+        fp32_hist = not self.p.deterministic_histogram and self.comm is None
+        key = (self.p.max_depth, id(self.qm), slot, fp32_hist)
         growers = getattr(self, "_device_growers", None)
         if growers is None:
             growers = {}
             self._device_growers = growers
         dg = growers.get(key)
         if dg is None:
-            dg = self.backend.DeviceGrower(self.state, self.p.max_depth)
+            dg = self.backend.DeviceGrower(self.state, self.p.max_depth, fp32_hist=fp32_hist)
             growers[key] = dg
         return dg
 
