@@ -13,11 +13,12 @@ sys.path.insert(0, os.path.join(os.path.dirname(__file__), ".."))
 
 from sagemaker_xgboost_container_amd.data.dmatrix import DeviceDMatrix  # noqa: E402
 from sagemaker_xgboost_container_amd.models import trainer  # noqa: E402
+from sagemaker_xgboost_container_amd.models.callback_api import TrainingCallback  # noqa: E402
 
 WARMUP, STEPS = 5, 30
 
 
-class _Timer:
+class _Timer(TrainingCallback):
     def __init__(self):
         self.rate = None
 
