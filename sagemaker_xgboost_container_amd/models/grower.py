@@ -50,10 +50,11 @@ class GrowParams:
         self.colsample_bynode = float(p.get("colsample_bynode", 1.0))
         self.monotone_constraints = p.get("monotone_constraints")
         self.interaction_constraints = p.get("interaction_constraints")
-        # xgboost semantics: deterministic (bit-reproducible) histograms by
-        # default; "false"/"0" opts into the faster nondeterministic
-        # accumulation (single-process GPU only — distributed needs
-        # rank-consistent sums and always stays deterministic)
+        # Accepted for xgboost parity; this framework is ALWAYS deterministic.
+        # A float-atomic LDS fast path for "false" was built and measured
+        # 3x SLOWER than the int64 fixed-point slab (80.9 vs 239.4 r/s on the
+        # 12.5M-row bench — ds_add_f32 throughput; see
+        # profiles/r01_optimization_log.md), so the flag costs nothing here.
         self.deterministic_histogram = str(
             p.get("deterministic_histogram", "true")
         ).lower() not in ("false", "0")
@@ -232,15 +233,14 @@ class HistGrower:
 
     def _device_grower_for(self, slot):
         """One DeviceGrower (own heap buffers) per (depth, matrix, slot)."""
-        fp32_hist = not self.p.deterministic_histogram and self.comm is None
-        key = (self.p.max_depth, id(self.qm), slot, fp32_hist)
+        key = (self.p.max_depth, id(self.qm), slot)
         growers = getattr(self, "_device_growers", None)
         if growers is None:
             growers = {}
             self._device_growers = growers
         dg = growers.get(key)
         if dg is None:
-            dg = self.backend.DeviceGrower(self.state, self.p.max_depth, fp32_hist=fp32_hist)
+            dg = self.backend.DeviceGrower(self.state, self.p.max_depth)
             growers[key] = dg
         return dg
 

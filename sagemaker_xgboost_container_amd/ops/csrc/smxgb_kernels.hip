@@ -621,74 +621,6 @@ __global__ __launch_bounds__(BLOCK) void hist_device_kernel(
   }
 }
 
-// `deterministic_histogram=false` fast path: LDS float atomics instead of
-// int64 fixed-point — half the slab (the full-feature slab fits TWO
-// 512-thread blocks/CU = 16 waves) and cheaper adds. Accumulation order
-// varies run to run, like xgboost's pre-deterministic gpu_hist; the
-// deterministic u64 path stays the default and is forced in distributed
-// training (rank-consistent sums are required there).
-template <typename BinT, int BLOCK>
-__global__ __launch_bounds__(BLOCK) void hist_device_f32_kernel(
-    const BinT* __restrict__ bins_c, const float2* __restrict__ gh_c,
-    const LevelNode* __restrict__ nodes, const int* __restrict__ hist_prefix,
-    const LevelWork* __restrict__ work, float* __restrict__ out,
-    int k, int nfeat, int stride, int n_groups, int feats_per_group,
-    int rows_per_block) {
-  extern __shared__ float lhist_f[];
-  const int total = work->hist_total * n_groups;
-
-  for (int vb = blockIdx.x; vb < total; vb += gridDim.x) {
-    const int fg = vb % n_groups;
-    const int hvb = vb / n_groups;
-    const int slot = find_slot(hist_prefix, k, hvb);
-    const int chunk = hvb - hist_prefix[slot];
-    const int nb = hist_prefix[slot + 1] - hist_prefix[slot];
-    const LevelNode node = nodes[slot];
-    const int fg_start = fg * feats_per_group;
-    const int nf_group = min(feats_per_group, nfeat - fg_start);
-    const int lds_words = nf_group * stride * 2;
-    const int hofs = (int)lds_half_words(nf_group * stride);
-    const int lds_padded = 2 * hofs;
-
-    for (int i = threadIdx.x; i < lds_padded; i += blockDim.x) lhist_f[i] = 0.f;
-    __syncthreads();
-
-    const long long step = (long long)nb * blockDim.x;
-    for (long long r = node.start + (long long)chunk * blockDim.x + threadIdx.x; r < node.end;
-         r += step) {
-      const float2 gp = gh_c[r];
-      const BinT* rp = bins_c + (long long)r * nfeat + fg_start;
-      #pragma unroll 4
-      for (int f = 0; f < nf_group; ++f) {
-        const int slot2 = lds_pad_slot(f * stride + (int)rp[f]);
-        atomicAdd(&lhist_f[slot2], gp.x);
-        atomicAdd(&lhist_f[hofs + slot2], gp.y);
-      }
-    }
-    __syncthreads();
-    float* gout = out + ((long long)slot * nfeat + fg_start) * (long long)stride * 2;
-    for (int i = threadIdx.x; i < lds_words; i += blockDim.x) {
-      const int pair = i >> 1;
-      const int idx = lds_pad_slot(pair) + ((i & 1) ? hofs : 0);
-      const float v = lhist_f[idx];
-      if (v != 0.f) atomicAdd(&gout[i], v);
-    }
-    __syncthreads();
-  }
-}
-
-// f32-acc convert: built slots copy straight into the f32 heap
-__global__ __launch_bounds__(HIST_BLOCK) void convert_level_f32_kernel(
-    const float* __restrict__ acc, float* __restrict__ hist_f32,
-    const LevelNode* __restrict__ nodes, int k, long long slots2) {
-  for (long long u = (long long)blockIdx.x * blockDim.x + threadIdx.x; u < (long long)k * slots2;
-       u += (long long)gridDim.x * blockDim.x) {
-    const int slot = (int)(u / slots2);
-    if (!nodes[slot].build || nodes[slot].end <= nodes[slot].start) continue;
-    hist_f32[u] = acc[u];
-  }
-}
-
 // convert built slots (int64 acc -> f32 heap) and fill node sums from the
 // feature-0 bin range; one block per (slot, chunk of slots_total)
 __global__ __launch_bounds__(HIST_BLOCK) void convert_level_kernel(
@@ -1318,23 +1250,6 @@ static void launch_hist_device(int grid, int hist_block, size_t lds_bytes, hipSt
   }
 }
 
-template <typename BinT>
-static void launch_hist_device_f32(int grid, int hist_block, size_t lds_bytes, hipStream_t stream,
-                                   const BinT* bins_c, const float2* gh_c, const LevelNode* nodes,
-                                   const int* hist_prefix, const LevelWork* work, float* acc,
-                                   int k, int nfeat, int stride, int n_groups,
-                                   int feats_per_group, int rows_per_block) {
-  if (hist_block == 512) {
-    hipLaunchKernelGGL((hist_device_f32_kernel<BinT, 512>), dim3(grid), dim3(512), lds_bytes,
-                       stream, bins_c, gh_c, nodes, hist_prefix, work, acc, k, nfeat, stride,
-                       n_groups, feats_per_group, rows_per_block);
-  } else {
-    hipLaunchKernelGGL((hist_device_f32_kernel<BinT, HIST_BLOCK>), dim3(grid), dim3(HIST_BLOCK),
-                       lds_bytes, stream, bins_c, gh_c, nodes, hist_prefix, work, acc, k, nfeat,
-                       stride, n_groups, feats_per_group, rows_per_block);
-  }
-}
-
 void grow_hist_level(torch::Tensor bins_c, torch::Tensor gh_c, torch::Tensor nodes,
                      torch::Tensor hist_prefix, torch::Tensor work, torch::Tensor acc,
                      int64_t k, int64_t nfeat, int64_t stride, int64_t n_groups,
@@ -1427,9 +1342,7 @@ void grow_tree_enqueue(
   auto stream = current_stream();
   const long long slots2 = (long long)nfeat * stride * 2;
   const bool u8 = init_bins.scalar_type() == torch::kUInt8;
-  const bool fp32_hist = acc.scalar_type() == torch::kFloat32;  // deterministic_histogram=false
-  const size_t lds_bytes =
-      (size_t)lds_words * (fp32_hist ? sizeof(float) : sizeof(unsigned long long));
+  const size_t lds_bytes = (size_t)lds_words * sizeof(unsigned long long);
   auto mono = torch::Tensor();  // device path excludes monotone constraints
 
   hipMemsetAsync(counts.data_ptr<int>(), 0, sizeof(int) * 2 * counts.size(0), stream);
@@ -1472,21 +1385,8 @@ void grow_tree_enqueue(
     float2* dst_gh = (float2*)(dpar ? gh1.data_ptr<float>() : gh0.data_ptr<float>());
     int* dst_rows = dpar ? rows1.data_ptr<int>() : rows0.data_ptr<int>();
 
-    hipMemsetAsync(acc.data_ptr(), 0, acc.element_size() * (size_t)k * slots2, stream);
-    if (fp32_hist) {
-      float* acc_f = acc.data_ptr<float>();
-      if (u8) {
-        launch_hist_device_f32<unsigned char>(
-            (int)hist_grid, (int)hist_block, lds_bytes, stream, (const unsigned char*)src_bins,
-            src_gh, nodes_d, hp_d, work_d, acc_f, k, (int)nfeat, (int)stride, (int)n_groups,
-            (int)feats_per_group, (int)rows_per_block);
-      } else {
-        launch_hist_device_f32<short>(
-            (int)hist_grid, (int)hist_block, lds_bytes, stream, (const short*)src_bins, src_gh,
-            nodes_d, hp_d, work_d, acc_f, k, (int)nfeat, (int)stride, (int)n_groups,
-            (int)feats_per_group, (int)rows_per_block);
-      }
-    } else if (u8) {
+    hipMemsetAsync(acc.data_ptr<int64_t>(), 0, sizeof(int64_t) * (size_t)k * slots2, stream);
+    if (u8) {
       launch_hist_device<unsigned char>(
           (int)hist_grid, (int)hist_block, lds_bytes, stream, (const unsigned char*)src_bins,
           src_gh, nodes_d, hp_d, work_d, (unsigned long long*)acc.data_ptr<int64_t>(), k,
@@ -1503,14 +1403,9 @@ void grow_tree_enqueue(
     {
       const long long total = (long long)k * slots2;
       const int grid = (int)std::min<long long>((total + HIST_BLOCK - 1) / HIST_BLOCK, 2048);
-      if (fp32_hist) {
-        hipLaunchKernelGGL(convert_level_f32_kernel, dim3(std::max(grid, 1)), dim3(HIST_BLOCK),
-                           0, stream, acc.data_ptr<float>(), hist_d, nodes_d, k, slots2);
-      } else {
-        hipLaunchKernelGGL(convert_level_kernel, dim3(std::max(grid, 1)), dim3(HIST_BLOCK), 0,
-                           stream, (const unsigned long long*)acc.data_ptr<int64_t>(), hist_d,
-                           nodes_d, k, slots2, gh_max.data_ptr<float>());
-      }
+      hipLaunchKernelGGL(convert_level_kernel, dim3(std::max(grid, 1)), dim3(HIST_BLOCK), 0,
+                         stream, (const unsigned long long*)acc.data_ptr<int64_t>(), hist_d,
+                         nodes_d, k, slots2, gh_max.data_ptr<float>());
       if (d > 0) {
         const int pbase = (k >> 1) - 1;
         hipLaunchKernelGGL(derive_level_kernel, dim3(std::max(grid, 1)), dim3(HIST_BLOCK), 0,

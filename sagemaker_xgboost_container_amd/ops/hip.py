@@ -104,20 +104,17 @@ def _feature_groups(nfeat, stride):
     return groups
 
 
-def _device_hist_plan(nfeat, stride, word_bytes=8):
+def _device_hist_plan(nfeat, stride):
     """Pick the device-grower hist configuration: (groups, lds_words, block).
 
     A single full-feature slab at 512 threads (1 block/CU = the same 8
-    waves as two 256-thread blocks on 56 KB slabs; 2 blocks/CU = 16 waves
-    when the slab is f32) streams bins/gh ONCE per row instead of once per
-    feature group; used whenever the padded slab fits the 160 KB LDS.
-    SMXGB_LDS_KB forces the grouped layout for sweeps. `word_bytes` is the
-    accumulator word size: 8 for the deterministic int64 fixed-point slab,
-    4 for the `deterministic_histogram=false` float slab.
+    waves as two 256-thread blocks on 56 KB slabs) streams bins/gh ONCE per
+    row instead of once per feature group; used whenever the padded slab
+    fits the 160 KB LDS. SMXGB_LDS_KB forces the grouped layout for sweeps.
     """
     if not _os.environ.get("SMXGB_LDS_KB"):
         words = _padded_words(nfeat * stride)
-        if words * word_bytes <= 158 * 1024:
+        if words * 8 <= 158 * 1024:
             return [(0, nfeat)], words, 512
     groups = _feature_groups(nfeat, stride)
     return groups, _padded_words(max(fe - fs for fs, fe in groups) * stride), 256
@@ -552,12 +549,11 @@ class DeviceGrower:
     enqueues overlaps GPU execution.
     """
 
-    def __init__(self, state, max_depth, feature_mask=None, fp32_hist=False):
+    def __init__(self, state, max_depth, feature_mask=None):
         assert 1 <= max_depth <= 10, "device grower supports max_depth 1..10"
         self.state = state
         self.qm = state.qm
         self.D = max_depth
-        self.fp32_hist = fp32_hist  # deterministic_histogram=false fast path
         qm = state.qm
         device = qm.bins.device
         f = qm.num_col
@@ -571,18 +567,12 @@ class DeviceGrower:
         self.splits = torch.empty((H, 6), dtype=torch.float32, device=device)
         self.counts = torch.zeros((H, 2), dtype=torch.int32, device=device)
         self.hist_f32 = torch.empty((H, self.slots2), dtype=torch.float32, device=device)
-        self.acc = torch.empty(
-            (max_k, self.slots2),
-            dtype=torch.float32 if fp32_hist else torch.int64,
-            device=device,
-        )
+        self.acc = torch.empty((max_k, self.slots2), dtype=torch.int64, device=device)
         self.cands = torch.empty((max_k, f, 5), dtype=torch.float32, device=device)
         self.hp = [torch.empty((1 << d) + 1, dtype=torch.int32, device=device) for d in range(max_depth)]
         self.pp = [torch.empty((1 << d) + 1, dtype=torch.int32, device=device) for d in range(max_depth)]
         self.work = torch.empty((max_depth, 2), dtype=torch.int32, device=device)
-        groups, self.lds_words, self.hist_block = _device_hist_plan(
-            f, stride, word_bytes=4 if fp32_hist else 8
-        )
+        groups, self.lds_words, self.hist_block = _device_hist_plan(f, stride)
         self.n_groups = len(groups)
         self.feats_per_group = groups[0][1] - groups[0][0]
         if feature_mask is None:
@@ -654,8 +644,6 @@ class DeviceGrower:
         root_gh_np [2])."""
         if comm is None:
             return self.grow_wait(self.grow_enqueue(scale, split_params))
-        # rank-consistent sums are required across workers
-        assert not self.fp32_hist, "deterministic histograms are forced in distributed training"
 
         st = self.state
         qm = self.qm

@@ -378,28 +378,22 @@ class TestMulticlassPipeline:
         np.testing.assert_allclose(p1, p2, rtol=1e-5, atol=1e-6)
 
 
-class TestNondeterministicHistogram:
-    """deterministic_histogram=false: float-atomic LDS accumulation. The
-    model may differ from the deterministic one on near-tie splits, so we
-    assert statistical equivalence, not bit equality."""
+class TestDeterministicHistogramFlag:
+    """The flag is accepted (xgboost parity) and training stays
+    deterministic either way — the measured float-atomic "fast" path was 3x
+    slower than the int64 slab, so both values run the same kernels."""
 
-    def test_statistically_equivalent(self):
+    def test_flag_accepted_and_identical(self):
         from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
         from sagemaker_xgboost_container_amd.models import trainer
 
         rng = np.random.default_rng(9)
-        X = rng.normal(size=(400_000, 16)).astype(np.float32)
-        y = (X[:, 0] + 0.5 * X[:, 3] - 0.2 * X[:, 7] > 0).astype(np.float32)
-        base = {"objective": "binary:logistic", "max_depth": 6, "eta": 0.3, "device": "cuda"}
+        X = rng.normal(size=(100_000, 12)).astype(np.float32)
+        y = (X[:, 0] + 0.5 * X[:, 3] > 0).astype(np.float32)
+        base = {"objective": "binary:logistic", "max_depth": 5, "eta": 0.3, "device": "cuda"}
 
-        res_det, res_fast = {}, {}
-        trainer.train(dict(base, deterministic_histogram="true"), DMatrix(X, label=y),
-                      num_boost_round=8, evals=[(DMatrix(X, label=y), "train")],
-                      evals_result=res_det, verbose_eval=False)
-        bst = trainer.train(dict(base, deterministic_histogram="false"), DMatrix(X, label=y),
-                            num_boost_round=8, evals=[(DMatrix(X, label=y), "train")],
-                            evals_result=res_fast, verbose_eval=False)
-        l_det = res_det["train"]["logloss"][-1]
-        l_fast = res_fast["train"]["logloss"][-1]
-        assert l_fast == pytest.approx(l_det, rel=0.05)
-        assert np.isfinite(bst.predict(X[:1000])).all()
+        b1 = trainer.train(dict(base, deterministic_histogram="true"), DMatrix(X, label=y),
+                           num_boost_round=4, verbose_eval=False)
+        b2 = trainer.train(dict(base, deterministic_histogram="false"), DMatrix(X, label=y),
+                           num_boost_round=4, verbose_eval=False)
+        np.testing.assert_array_equal(b1.predict(X[:2000]), b2.predict(X[:2000]))
