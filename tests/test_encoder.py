@@ -96,3 +96,31 @@ class TestDecodeDispatch:
     def test_unsupported_type(self):
         with pytest.raises(exc.UserError, match="Unsupported content type"):
             encoder.decode(b"x", "application/pdf")
+
+
+class TestServingEncoders:
+    """utils/serving_encoders.py: response body encoders by accept type."""
+
+    def test_array_to_csv(self):
+        from sagemaker_xgboost_container_amd.utils import serving_encoders
+
+        out = serving_encoders.array_to_csv([[1.0, 2.0], [3.0, 4.0]])
+        assert out.strip().split("\n") == ["1.0,2.0", "3.0,4.0"]
+
+    def test_array_to_json(self):
+        from sagemaker_xgboost_container_amd.utils import serving_encoders
+
+        assert json.loads(serving_encoders.array_to_json([0.25, 0.75])) == [0.25, 0.75]
+
+    def test_npy_roundtrip(self):
+        from sagemaker_xgboost_container_amd.utils import serving_encoders
+
+        payload = serving_encoders.encode(np.array([1.5, 2.5]), "application/x-npy")
+        out = np.load(io.BytesIO(payload), allow_pickle=False)
+        np.testing.assert_allclose(out, [1.5, 2.5])
+
+    def test_unsupported_accept(self):
+        from sagemaker_xgboost_container_amd.utils import serving_encoders
+
+        with pytest.raises(Exception):
+            serving_encoders.encode([1.0], "application/pdf")
