@@ -352,27 +352,35 @@ def train(
                         es.margin = booster._margin(es.X).reshape(es.X.shape[0], -1).clone()
 
         # -- evaluation ----------------------------------------------------
-        # Each rank evaluates its local shard; values are then aggregated to
-        # one identical number on every rank (ONE fused allreduce — exact for
-        # ratio-of-sums metrics, per-worker weighted average for rank metrics;
-        # the reference gets the same property from rabit inside xgboost).
+        # Each rank evaluates its local shard; native metrics are then
+        # aggregated to one identical number on every rank (ONE fused
+        # allreduce — exact for ratio-of-sums metrics, per-worker weighted
+        # average for rank metrics; the reference gets the same property
+        # from rabit inside xgboost). Custom feval metrics stay rank-local,
+        # exactly as xgboost's feval does in distributed mode; an empty
+        # shard contributes (0 value, 0 mass) so it can't NaN the sums.
         results = []
         masses = []
+        feval_results = []
         for es in eval_sets:
             es_margin = margin if es.is_train else es.margin
             m = es_margin.squeeze(1) if n_outputs == 1 else es_margin
+            empty = es.y.numel() == 0
             for metric_name in metric_names:
+                if empty:
+                    results.append((es.name, metric_name, 0.0))
+                    masses.append(0.0)
+                    continue
                 value = evaluate_metric(metric_name, m, es.y, es.w, objective)
                 results.append((es.name, metric_name, value))
                 masses.append(eval_metrics.metric_mass(metric_name, es.y, es.w, objective))
-            if feval is not None:
+            if feval is not None and not empty:
                 m_np = m.cpu().numpy()
                 custom = feval(m_np, es.dmatrix)
                 if isinstance(custom, tuple):
                     custom = [custom]
                 for metric_name, value in custom:
-                    results.append((es.name, metric_name, float(value)))
-                    masses.append(float(es.y.numel()))
+                    feval_results.append((es.name, metric_name, float(value)))
 
         if comm is not None and comm.world_size > 1 and results:
             vm = torch.tensor(
@@ -387,6 +395,7 @@ def train(
                 (ds, nm, eval_metrics.from_agg_space(nm, float(vm[i] / vm[k + i]) if float(vm[k + i]) > 0 else 0.0))
                 for i, (ds, nm, _) in enumerate(results)
             ]
+        results.extend(feval_results)
 
         if container.after_iteration(booster, epoch, results):
             break
