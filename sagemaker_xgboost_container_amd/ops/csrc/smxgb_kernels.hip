@@ -595,12 +595,36 @@ __global__ __launch_bounds__(BLOCK) void hist_device_kernel(
     __syncthreads();
 
     const long long step = (long long)nb * blockDim.x;
+    const bool vec4 = sizeof(BinT) == 1 && (nf_group & 3) == 0 && (nfeat & 3) == 0 &&
+                      (fg_start & 3) == 0;
     for (long long r = node.start + (long long)chunk * blockDim.x + threadIdx.x; r < node.end;
          r += step) {
       const float2 gp = gh_c[r];
       const unsigned long long gfix = (unsigned long long)(long long)llrintf(gp.x * scale_g);
       const unsigned long long hfix = (unsigned long long)(long long)llrintf(gp.y * scale_h);
       const BinT* rp = bins_c + (long long)r * nfeat + fg_start;
+      if (vec4) {
+        // 4 bins per dword load (rows are 4-aligned when nfeat % 4 == 0)
+        const uchar4* rp4 = reinterpret_cast<const uchar4*>(rp);
+        #pragma unroll 2
+        for (int f4 = 0; f4 < (nf_group >> 2); ++f4) {
+          const uchar4 b4 = rp4[f4];
+          const int base = (f4 << 2) * stride;
+          const int s0 = lds_pad_slot(base + (int)b4.x);
+          const int s1 = lds_pad_slot(base + stride + (int)b4.y);
+          const int s2 = lds_pad_slot(base + 2 * stride + (int)b4.z);
+          const int s3 = lds_pad_slot(base + 3 * stride + (int)b4.w);
+          atomicAdd(&lhist[s0], gfix);
+          atomicAdd(&lhist[hofs + s0], hfix);
+          atomicAdd(&lhist[s1], gfix);
+          atomicAdd(&lhist[hofs + s1], hfix);
+          atomicAdd(&lhist[s2], gfix);
+          atomicAdd(&lhist[hofs + s2], hfix);
+          atomicAdd(&lhist[s3], gfix);
+          atomicAdd(&lhist[hofs + s3], hfix);
+        }
+        continue;
+      }
       #pragma unroll 4
       for (int f = 0; f < nf_group; ++f) {
         const int slot2 = lds_pad_slot(f * stride + (int)rp[f]);
