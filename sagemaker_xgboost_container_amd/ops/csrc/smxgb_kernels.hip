@@ -597,14 +597,61 @@ __global__ __launch_bounds__(BLOCK) void hist_device_kernel(
     const long long step = (long long)nb * blockDim.x;
     const bool vec4 = sizeof(BinT) == 1 && (nf_group & 3) == 0 && (nfeat & 3) == 0 &&
                       (fg_start & 3) == 0;
-    for (long long r = node.start + (long long)chunk * blockDim.x + threadIdx.x; r < node.end;
-         r += step) {
+    long long r = node.start + (long long)chunk * blockDim.x + threadIdx.x;
+    if (vec4) {
+      // 4 bins per dword load (rows are 4-aligned when nfeat % 4 == 0);
+      // 2 rows in flight per iteration so the gathers overlap
+      const int nd = nf_group >> 2;
+      for (; r + step < node.end; r += 2 * step) {
+        const long long r2 = r + step;
+        const float2 gp0 = gh_c[r];
+        const float2 gp1 = gh_c[r2];
+        const uchar4* rp0 = reinterpret_cast<const uchar4*>(
+            bins_c + (long long)r * nfeat + fg_start);
+        const uchar4* rp1 = reinterpret_cast<const uchar4*>(
+            bins_c + r2 * nfeat + fg_start);
+        const unsigned long long g0 = (unsigned long long)(long long)llrintf(gp0.x * scale_g);
+        const unsigned long long h0 = (unsigned long long)(long long)llrintf(gp0.y * scale_h);
+        const unsigned long long g1 = (unsigned long long)(long long)llrintf(gp1.x * scale_g);
+        const unsigned long long h1 = (unsigned long long)(long long)llrintf(gp1.y * scale_h);
+        #pragma unroll 2
+        for (int f4 = 0; f4 < nd; ++f4) {
+          const uchar4 a = rp0[f4];
+          const uchar4 b = rp1[f4];
+          const int base = (f4 << 2) * stride;
+          const int a0 = lds_pad_slot(base + (int)a.x);
+          const int a1 = lds_pad_slot(base + stride + (int)a.y);
+          const int a2 = lds_pad_slot(base + 2 * stride + (int)a.z);
+          const int a3 = lds_pad_slot(base + 3 * stride + (int)a.w);
+          const int b0 = lds_pad_slot(base + (int)b.x);
+          const int b1 = lds_pad_slot(base + stride + (int)b.y);
+          const int b2 = lds_pad_slot(base + 2 * stride + (int)b.z);
+          const int b3 = lds_pad_slot(base + 3 * stride + (int)b.w);
+          atomicAdd(&lhist[a0], g0);
+          atomicAdd(&lhist[hofs + a0], h0);
+          atomicAdd(&lhist[b0], g1);
+          atomicAdd(&lhist[hofs + b0], h1);
+          atomicAdd(&lhist[a1], g0);
+          atomicAdd(&lhist[hofs + a1], h0);
+          atomicAdd(&lhist[b1], g1);
+          atomicAdd(&lhist[hofs + b1], h1);
+          atomicAdd(&lhist[a2], g0);
+          atomicAdd(&lhist[hofs + a2], h0);
+          atomicAdd(&lhist[b2], g1);
+          atomicAdd(&lhist[hofs + b2], h1);
+          atomicAdd(&lhist[a3], g0);
+          atomicAdd(&lhist[hofs + a3], h0);
+          atomicAdd(&lhist[b3], g1);
+          atomicAdd(&lhist[hofs + b3], h1);
+        }
+      }
+    }
+    for (; r < node.end; r += step) {
       const float2 gp = gh_c[r];
       const unsigned long long gfix = (unsigned long long)(long long)llrintf(gp.x * scale_g);
       const unsigned long long hfix = (unsigned long long)(long long)llrintf(gp.y * scale_h);
       const BinT* rp = bins_c + (long long)r * nfeat + fg_start;
       if (vec4) {
-        // 4 bins per dword load (rows are 4-aligned when nfeat % 4 == 0)
         const uchar4* rp4 = reinterpret_cast<const uchar4*>(rp);
         #pragma unroll 2
         for (int f4 = 0; f4 < (nf_group >> 2); ++f4) {
