@@ -600,49 +600,41 @@ __global__ __launch_bounds__(BLOCK) void hist_device_kernel(
     long long r = node.start + (long long)chunk * blockDim.x + threadIdx.x;
     if (vec4) {
       // 4 bins per dword load (rows are 4-aligned when nfeat % 4 == 0);
-      // 2 rows in flight per iteration so the gathers overlap
+      // HROWS rows in flight per iteration so the gathers overlap
+      constexpr int HROWS = 4;
       const int nd = nf_group >> 2;
-      for (; r + step < node.end; r += 2 * step) {
-        const long long r2 = r + step;
-        const float2 gp0 = gh_c[r];
-        const float2 gp1 = gh_c[r2];
-        const uchar4* rp0 = reinterpret_cast<const uchar4*>(
-            bins_c + (long long)r * nfeat + fg_start);
-        const uchar4* rp1 = reinterpret_cast<const uchar4*>(
-            bins_c + r2 * nfeat + fg_start);
-        const unsigned long long g0 = (unsigned long long)(long long)llrintf(gp0.x * scale_g);
-        const unsigned long long h0 = (unsigned long long)(long long)llrintf(gp0.y * scale_h);
-        const unsigned long long g1 = (unsigned long long)(long long)llrintf(gp1.x * scale_g);
-        const unsigned long long h1 = (unsigned long long)(long long)llrintf(gp1.y * scale_h);
+      for (; r + (HROWS - 1) * step < node.end; r += HROWS * step) {
+        unsigned long long gf[HROWS], hf[HROWS];
+        const uchar4* rp[HROWS];
+        #pragma unroll
+        for (int u = 0; u < HROWS; ++u) {
+          const long long ru = r + u * step;
+          const float2 gp = gh_c[ru];
+          gf[u] = (unsigned long long)(long long)llrintf(gp.x * scale_g);
+          hf[u] = (unsigned long long)(long long)llrintf(gp.y * scale_h);
+          rp[u] = reinterpret_cast<const uchar4*>(bins_c + ru * nfeat + fg_start);
+        }
         #pragma unroll 2
         for (int f4 = 0; f4 < nd; ++f4) {
-          const uchar4 a = rp0[f4];
-          const uchar4 b = rp1[f4];
           const int base = (f4 << 2) * stride;
-          const int a0 = lds_pad_slot(base + (int)a.x);
-          const int a1 = lds_pad_slot(base + stride + (int)a.y);
-          const int a2 = lds_pad_slot(base + 2 * stride + (int)a.z);
-          const int a3 = lds_pad_slot(base + 3 * stride + (int)a.w);
-          const int b0 = lds_pad_slot(base + (int)b.x);
-          const int b1 = lds_pad_slot(base + stride + (int)b.y);
-          const int b2 = lds_pad_slot(base + 2 * stride + (int)b.z);
-          const int b3 = lds_pad_slot(base + 3 * stride + (int)b.w);
-          atomicAdd(&lhist[a0], g0);
-          atomicAdd(&lhist[hofs + a0], h0);
-          atomicAdd(&lhist[b0], g1);
-          atomicAdd(&lhist[hofs + b0], h1);
-          atomicAdd(&lhist[a1], g0);
-          atomicAdd(&lhist[hofs + a1], h0);
-          atomicAdd(&lhist[b1], g1);
-          atomicAdd(&lhist[hofs + b1], h1);
-          atomicAdd(&lhist[a2], g0);
-          atomicAdd(&lhist[hofs + a2], h0);
-          atomicAdd(&lhist[b2], g1);
-          atomicAdd(&lhist[hofs + b2], h1);
-          atomicAdd(&lhist[a3], g0);
-          atomicAdd(&lhist[hofs + a3], h0);
-          atomicAdd(&lhist[b3], g1);
-          atomicAdd(&lhist[hofs + b3], h1);
+          uchar4 b4[HROWS];
+          #pragma unroll
+          for (int u = 0; u < HROWS; ++u) b4[u] = rp[u][f4];
+          #pragma unroll
+          for (int u = 0; u < HROWS; ++u) {
+            const int s0 = lds_pad_slot(base + (int)b4[u].x);
+            const int s1 = lds_pad_slot(base + stride + (int)b4[u].y);
+            const int s2 = lds_pad_slot(base + 2 * stride + (int)b4[u].z);
+            const int s3 = lds_pad_slot(base + 3 * stride + (int)b4[u].w);
+            atomicAdd(&lhist[s0], gf[u]);
+            atomicAdd(&lhist[hofs + s0], hf[u]);
+            atomicAdd(&lhist[s1], gf[u]);
+            atomicAdd(&lhist[hofs + s1], hf[u]);
+            atomicAdd(&lhist[s2], gf[u]);
+            atomicAdd(&lhist[hofs + s2], hf[u]);
+            atomicAdd(&lhist[s3], gf[u]);
+            atomicAdd(&lhist[hofs + s3], hf[u]);
+          }
         }
       }
     }
