@@ -116,25 +116,26 @@ def make_cuts_distributed(X, max_bin=256, sample_weight=None, comm=None, n_candi
 
     cand = torch.full((f, ncand), float("nan"), device=device)
     mass = torch.zeros(f, device=device)
+    # one batched column sort (NaNs land at the bottom), as in make_cuts
+    sorted_all, order_all = torch.sort(X, dim=0)
+    finite_counts = (~torch.isnan(X)).sum(0).tolist()
     for j in range(f):
-        col = X[:, j]
-        finite_mask = ~torch.isnan(col)
-        finite = col[finite_mask]
-        if finite.numel() == 0:
+        m = int(finite_counts[j])
+        if m == 0:
             continue
-        sorted_vals, order = torch.sort(finite)
+        sorted_vals = sorted_all[:m, j]
         if sample_weight is not None:
-            w = sample_weight[finite_mask][order]
+            w = sample_weight[order_all[:m, j]]
             cw = torch.cumsum(w, 0)
             total = cw[-1]
             targets = (torch.arange(ncand, device=device, dtype=torch.float32) + 0.5) / ncand * total
-            pos = torch.searchsorted(cw, targets).clamp_(0, finite.numel() - 1)
+            pos = torch.searchsorted(cw, targets).clamp_(0, m - 1)
             cand[j] = sorted_vals[pos]
             mass[j] = total
         else:
-            pos = ((torch.arange(ncand, device=device, dtype=torch.float64) + 0.5) / ncand * finite.numel()).long()
-            cand[j] = sorted_vals[pos.clamp_(0, finite.numel() - 1)]
-            mass[j] = float(finite.numel())
+            pos = ((torch.arange(ncand, device=device, dtype=torch.float64) + 0.5) / ncand * m).long()
+            cand[j] = sorted_vals[pos.clamp_(0, m - 1)]
+            mass[j] = float(m)
 
     if comm is not None and comm.world_size > 1:
         world = comm.world_size
