@@ -251,3 +251,41 @@ class TestUserModuleServing:
         r = client.post("/invocations", content=b"0.1,0.2,0.3,0.4",
                         headers={"Content-Type": "text/csv", "Accept": "text/csv"})
         assert r.status_code == 200
+
+
+class TestMalformedPayloadFuzz:
+    """Random/malformed bodies must map to 4xx client errors — never a 500
+    or a worker crash (the reference maps decode failures the same way)."""
+
+    @pytest.mark.parametrize("payload,ctype", [
+        (b"\x00\xff\xfe\x01\x02", "text/csv"),
+        (b"not,numbers,at,all\nx,y,z,w", "text/csv"),
+        (b"1:abc 2:def", "text/libsvm"),
+        (b"\x93NUMPY garbage", "application/x-npy"),
+        (b"{\"json\": \"not a dmatrix\"}", "application/json"),
+        (b"1,2\n1,2,3,4,5,6,7,8,9", "text/csv"),  # ragged + wrong width
+        (b"\xed\xa0\x80 invalid utf8 continuation", "text/csv"),
+    ])
+    def test_bad_bodies_are_client_errors(self, client, payload, ctype):
+        r = client.post("/invocations", content=payload,
+                        headers={"Content-Type": ctype})
+        assert 400 <= r.status_code < 500, (payload, ctype, r.status_code, r.text[:200])
+
+    def test_empty_body_is_204(self, client):
+        # reference contract: NoContentInferenceError -> 204, not a 4xx
+        r = client.post("/invocations", content=b"", headers={"Content-Type": "text/csv"})
+        assert r.status_code == 204
+
+    def test_hypothesis_fuzz_csv(self, client):
+        from hypothesis import HealthCheck, given, settings
+        from hypothesis import strategies as st
+
+        @settings(max_examples=30, deadline=None,
+                  suppress_health_check=[HealthCheck.function_scoped_fixture])
+        @given(body=st.binary(max_size=256))
+        def run(body):
+            r = client.post("/invocations", content=body,
+                            headers={"Content-Type": "text/csv"})
+            assert r.status_code < 500, (body, r.status_code)
+
+        run()
