@@ -289,3 +289,30 @@ class TestMalformedPayloadFuzz:
             assert r.status_code < 500, (body, r.status_code)
 
         run()
+
+
+class TestMmePayloadCap:
+    def test_oversize_invoke_413(self, tmp_path, monkeypatch):
+        from sagemaker_xgboost_container_amd import serving_mms
+
+        monkeypatch.setenv("SAGEMAKER_MAX_PAYLOAD_IN_MB", "1")
+        bst = _train_booster()
+        model_a = tmp_path / "m"
+        model_a.mkdir()
+        bst.save_model(model_a / "xgboost-model")
+        client = TestClient(serving_mms.app)
+        assert client.post("/models", json={"model_name": "big", "url": str(model_a)}).status_code == 200
+        try:
+            big = b"1,2,3,4\n" * (1024 * 1024 // 4)  # ~2 MB > 1 MB cap
+            r = client.post("/models/big/invoke", content=big,
+                            headers={"Content-Type": "text/csv"})
+            assert r.status_code == 413
+        finally:
+            client.delete("/models/big")
+
+    def test_cap_above_20mb_rejected(self, monkeypatch):
+        from sagemaker_xgboost_container_amd import serving_mms
+
+        monkeypatch.setenv("SAGEMAKER_MAX_PAYLOAD_IN_MB", "21")
+        with pytest.raises(ValueError, match="cannot exceed 20"):
+            serving_mms._max_payload_bytes()
