@@ -162,3 +162,26 @@ class TestMetricsAndExceptions:
         assert "boom" in e.message and "ValueError" in e.message
         e2 = exc.AlgorithmError(caused_by=ValueError("inner"))
         assert "inner" in e2.message
+
+
+class TestIntegrationLogging:
+    """`[ts:LEVEL] msg` console format (reference integration.py:16-52) —
+    the format CloudWatch scrapes eval lines from."""
+
+    def test_format_string(self):
+        import logging
+        import re
+
+        from sagemaker_xgboost_container_amd.algorithm_mode import integration
+
+        fmt = integration.LOGGING_CONFIG["formatters"]["standard"]
+        record = logging.LogRecord("x", logging.INFO, "f.py", 1, "hello eval", None, None)
+        formatter = logging.Formatter(fmt["format"], datefmt=fmt.get("datefmt"))
+        line = formatter.format(record)
+        assert re.match(r"^\[\d{4}-\d{2}-\d{2}:\d{2}:\d{2}:\d{2}:INFO\] hello eval$", line), line
+
+    def test_setup_returns_named_logger(self):
+        from sagemaker_xgboost_container_amd.algorithm_mode import integration
+
+        lg = integration.setup_main_logger("smoke-test-logger")
+        assert lg.name == "smoke-test-logger"
