@@ -192,3 +192,20 @@ class TestDMatrixCore:
         f.write_text("1:0.7 0:1.0\n0:0.3 1:2.0\n")
         dm = DMatrix(f"{f}?format=libsvm")
         np.testing.assert_allclose(dm.get_weight(), [0.7, 0.3])
+
+
+def test_folder_depth_limit_raises(tmp_path):
+    """Channel dirs nested past MAX_FOLDER_DEPTH raise a UserError
+    (reference data_utils.py:476-545 staging contract)."""
+    from sagemaker_xgboost_container_amd.data import data_utils as du
+    from sagemaker_xgboost_container_amd.toolkit import exceptions as exc
+
+    deep = tmp_path
+    for i in range(du.MAX_FOLDER_DEPTH + 2):
+        deep = deep / f"level{i}"
+        deep.mkdir()
+    (deep / "data.csv").write_text("1,2,3\n")
+    dest = tmp_path / "staged"
+    dest.mkdir()
+    with pytest.raises(exc.UserError, match="depth"):
+        du._stage_folder(str(dest), str(tmp_path / "level0"), 1)
