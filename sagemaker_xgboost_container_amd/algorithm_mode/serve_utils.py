@@ -38,6 +38,7 @@ from ..data.dmatrix import DMatrix
 from ..data.encoder import json_to_jsonlines
 from ..data.recordio_protobuf import write_recordio_protobuf
 from ..models.booster import Booster
+from ..models.legacy_binary import load_pickled_booster
 
 import logging
 
@@ -157,7 +158,10 @@ def get_loaded_booster(model_dir, ensemble=False):
         logging.info("Loading the model from %s", full_model_path)
         try:
             with open(full_model_path, "rb") as f:
-                booster = pkl.load(f)
+                # handles pickles of the native Booster AND of upstream
+                # xgboost.core.Booster (reference serve_utils.py:180-182 is
+                # pickle-first; prior-container models arrive this way)
+                booster = load_pickled_booster(f.read())
             if not isinstance(booster, Booster):
                 raise TypeError(f"Pickled object is {type(booster)}, not a Booster")
             model_format = PKL_FORMAT

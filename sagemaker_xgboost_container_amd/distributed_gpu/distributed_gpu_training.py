@@ -86,13 +86,24 @@ def _worker(local_rank, num_gpus, sm_hosts, current_host, hyperparameters, train
     shard_rows = np.arange(rank, train_dmatrix.num_row(), world_size)
     shard = train_dmatrix.slice(shard_rows)
 
+    # Validation is row-sharded too: every rank gets an eval set (possibly
+    # 0 rows on small data), so eval-set/callback structure is identical
+    # across ranks — the fused metric allreduce needs matching tensor
+    # lengths, and EarlyStopping must fire on the same (globally
+    # aggregated, hence identical) metric on every rank or the collective
+    # deadlocks. A 0-row shard contributes (0 value, 0 mass) to the
+    # aggregation, so global metrics stay exact.
+    if val_dmatrix is not None:
+        val_rows = np.arange(rank, val_dmatrix.num_row(), world_size)
+        val_dmatrix = val_dmatrix.slice(val_rows)
+
     hp = dict(hyperparameters)
     hp["tree_method"] = GPU_TREE_METHOD
     hp["device"] = f"cuda:{local_rank}"
     train_job(
         train_cfg=hp,
         train_dmatrix=shard,
-        val_dmatrix=val_dmatrix if rank == 0 else None,
+        val_dmatrix=val_dmatrix,
         train_val_dmatrix=shard,
         model_dir=model_dir,
         checkpoint_dir=checkpoint_dir,

@@ -22,6 +22,39 @@ from .tree import Tree
 MODEL_VERSION = [3, 0, 5]  # schema-compatible xgboost version
 
 
+def objective_params_from_json(obj_json, params):
+    """Restore hyperparameters from a model/config objective block into
+    `params` (the inverse of Booster._objective_json). Shared by
+    load_json, save_config round-trips and the legacy-binary loader."""
+    blocks = {
+        "reg_loss_param": (("scale_pos_weight", float),),
+        "softmax_multiclass_param": (("num_class", int),),
+        "poisson_regression_param": (("max_delta_step", float),),
+        "tweedie_regression_param": (("tweedie_variance_power", float),),
+        "pseudo_huber_param": (("huber_slope", float),),
+        "aft_loss_param": (
+            ("aft_loss_distribution", str),
+            ("aft_loss_distribution_scale", float),
+        ),
+        "lambdarank_param": (
+            ("lambdarank_num_pair_per_sample", int),
+            ("lambdarank_pair_method", str),
+            ("lambdarank_unbiased", int),
+        ),
+    }
+    for block_name, fields in blocks.items():
+        block = obj_json.get(block_name)
+        if not isinstance(block, dict):
+            continue
+        for field, conv in fields:
+            if field in block:
+                try:
+                    params[field] = conv(block[field])
+                except (TypeError, ValueError):
+                    pass
+    return params
+
+
 class Booster:
     def __init__(self, params=None, num_features=0, feature_names=None):
         self.params = dict(params or {})
@@ -339,18 +372,42 @@ class Booster:
         tree.sum_hess = np.asarray(obj.get("sum_hessian", np.zeros(len(left))), dtype=np.float32)
         return tree
 
+    # objectives whose upstream SaveConfig writes a reg_loss_param block
+    _REG_LOSS_OBJECTIVES = (
+        "reg:squarederror", "reg:linear", "reg:squaredlogerror", "reg:logistic",
+        "binary:logistic", "binary:logitraw",
+    )
+
     def _objective_json(self):
+        """Objective block matching what upstream xgboost's SaveConfig
+        writes for every objective family — the loader there reads these
+        param sub-objects unconditionally, so each family must emit its
+        block (advisor round-1 finding)."""
         name = self.objective_name
         obj = {"name": name}
-        if name.startswith("binary:") or name == "reg:logistic":
-            obj["reg_loss_param"] = {"scale_pos_weight": str(self.params.get("scale_pos_weight", 1.0))}
+        p = self.params
+        if name in self._REG_LOSS_OBJECTIVES:
+            obj["reg_loss_param"] = {"scale_pos_weight": str(p.get("scale_pos_weight", 1.0))}
         elif name.startswith("multi:"):
             obj["softmax_multiclass_param"] = {"num_class": str(self.num_class)}
         elif name == "count:poisson":
-            obj["poisson_regression_param"] = {"max_delta_step": str(self.params.get("max_delta_step", 0.7))}
+            obj["poisson_regression_param"] = {"max_delta_step": str(p.get("max_delta_step", 0.7))}
         elif name == "reg:tweedie":
             obj["tweedie_regression_param"] = {
-                "tweedie_variance_power": str(self.params.get("tweedie_variance_power", 1.5))
+                "tweedie_variance_power": str(p.get("tweedie_variance_power", 1.5))
+            }
+        elif name == "reg:pseudohubererror":
+            obj["pseudo_huber_param"] = {"huber_slope": str(p.get("huber_slope", 1.0))}
+        elif name == "survival:aft":
+            obj["aft_loss_param"] = {
+                "aft_loss_distribution": str(p.get("aft_loss_distribution", "normal")),
+                "aft_loss_distribution_scale": str(p.get("aft_loss_distribution_scale", 1.0)),
+            }
+        elif name.startswith("rank:"):
+            obj["lambdarank_param"] = {
+                "lambdarank_num_pair_per_sample": str(p.get("lambdarank_num_pair_per_sample", 1)),
+                "lambdarank_pair_method": str(p.get("lambdarank_pair_method", "mean")),
+                "lambdarank_unbiased": str(p.get("lambdarank_unbiased", 0)),
             }
         return obj
 
@@ -413,6 +470,9 @@ class Booster:
         gb = learner["gradient_booster"]["model"]
         lmp = learner["learner_model_param"]
         self.params["objective"] = learner["objective"]["name"]
+        # restore objective parameters stored in the model (upstream does;
+        # without this, warm-start/predict silently uses defaults)
+        objective_params_from_json(learner.get("objective", {}), self.params)
         if int(lmp.get("num_class", "0") or 0) > 0:
             self.params["num_class"] = int(lmp["num_class"])
         self.params["base_score"] = float(lmp.get("base_score", 0.5))
@@ -454,7 +514,9 @@ class Booster:
         return self
 
     def load_model(self, path):
-        """Load a Booster file: JSON or UBJSON (xgboost >= 1.6 default)."""
+        """Load a Booster file: JSON, UBJSON (xgboost >= 1.6 default), or
+        the deprecated binary format older reference containers produced
+        (reference serve_utils.py:184-186 loads the same three)."""
         with open(path, "rb") as f:
             raw = f.read()
         head = raw[:1]
@@ -467,7 +529,13 @@ class Booster:
             from ..utils import ubjson
 
             return self.load_json(ubjson.loads(raw))
-        raise ValueError(f"Unsupported model format in {path} (expected JSON/UBJSON Booster)")
+        from .legacy_binary import looks_like_legacy_binary, parse_legacy_binary
+
+        if looks_like_legacy_binary(raw):
+            return parse_legacy_binary(raw, booster=self)
+        raise ValueError(
+            f"Unsupported model format in {path} (expected JSON/UBJSON/legacy-binary Booster)"
+        )
 
     def save_model_ubj(self, path):
         from ..utils import ubjson

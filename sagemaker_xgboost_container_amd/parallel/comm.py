@@ -62,7 +62,13 @@ def init_from_env(backend=None, timeout_s=1800):
         return Communicator()
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
     if world_size <= 1:
-        return None
+        # Launched under torchrun with one rank: still build the process
+        # group so the nccl(=RCCL) collective path is exercised end to end
+        # (1-rank allreduces run real RCCL kernels — the rehearsal the
+        # 1-GPU box can do). Plain `python bench.py` has no RANK and stays
+        # communicator-free.
+        if "RANK" not in os.environ or os.environ.get("SMXGB_FORCE_SINGLE") == "1":
+            return None
     if backend is None:
         backend = os.environ.get("SMXGB_COMM_BACKEND") or (
             "nccl" if torch.cuda.is_available() else "gloo"
