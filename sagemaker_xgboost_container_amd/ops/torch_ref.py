@@ -250,8 +250,12 @@ def make_flat_forest(trees, tree_info, weight_drop, device):
         "threshold": torch.from_numpy(np.concatenate([t.threshold for t in trees]).astype(np.float32)).to(device),
         "default_left": torch.from_numpy(np.concatenate([t.default_left for t in trees])).to(device),
         "value": torch.from_numpy(value).to(device),
+        "cover": torch.from_numpy(
+            np.concatenate([np.asarray(t.sum_hess, dtype=np.float32) for t in trees])
+        ).to(device),
         "tree_root": torch.from_numpy(offsets[:-1].astype(np.int64)).to(device),
         "tree_cls": torch.from_numpy(np.asarray(tree_info, dtype=np.int64)).to(device),
+        "max_depth": max((t.max_depth() for t in trees), default=0),
         "n_trees": len(trees),
     }
 
@@ -362,3 +366,199 @@ class TreeState:
 
 def make_tree_state(qm, gh, sample_rows=None, slot=0):
     return TreeState(qm, gh, sample_rows)  # CPU reference: slot unused
+
+
+# ---------------------------------------------------------------------------
+# Exact TreeSHAP + pred_leaf (host). C++ fast path (tree_shap_cpu /
+# pred_leaf_cpu in text_parsers.cpp); pure-Python recursion as the
+# no-extension fallback. Replaces the reference's native pred_contribs
+# (booster.predict(pred_contribs=True), reference test_abalone.py:65).
+# ---------------------------------------------------------------------------
+def _flat_i32(flat):
+    if "_i32" not in flat:
+        flat["_i32"] = {
+            "left": flat["left"].to(torch.int32).contiguous(),
+            "right": flat["right"].to(torch.int32).contiguous(),
+            "feature": flat["feature"].to(torch.int32).contiguous(),
+            "default_left": flat["default_left"].to(torch.uint8).contiguous(),
+            "tree_root": flat["tree_root"].to(torch.int32).contiguous(),
+            "tree_cls": flat["tree_cls"].to(torch.int32).contiguous(),
+        }
+    return flat["_i32"]
+
+
+def _py_tree_shap_one(flat, xr, phi, root):
+    """Pure-Python exact TreeSHAP for one (row, tree) — fallback/oracle."""
+    import math
+
+    left = flat["left"].numpy()
+    right = flat["right"].numpy()
+    feat = flat["feature"].numpy()
+    thresh = flat["threshold"].numpy()
+    defl = flat["default_left"].numpy()
+    value = flat["value"].numpy()
+    cover = flat["cover"].numpy()
+
+    def extend(path, zf, of, fi):
+        # deep-copy elements: both recursion branches extend the same
+        # parent path and must not share mutable state
+        path = [list(p) for p in path] + [[fi, zf, of, 1.0 if not path else 0.0]]
+        d = len(path) - 1
+        for i in range(d - 1, -1, -1):
+            path[i + 1][3] += of * path[i][3] * (i + 1) / (d + 1)
+            path[i][3] = zf * path[i][3] * (d - i) / (d + 1)
+        return path
+
+    def unwind(path, i):
+        d = len(path) - 1
+        of, zf = path[i][2], path[i][1]
+        out = [list(p) for p in path[:d]]
+        nxt = path[d][3]
+        for j in range(d - 1, -1, -1):
+            if of != 0:
+                tmp = out[j][3]
+                out[j][3] = nxt * (d + 1) / ((j + 1) * of)
+                nxt = tmp - out[j][3] * zf * (d - j) / (d + 1)
+            else:
+                out[j][3] = out[j][3] * (d + 1) / (zf * (d - j))
+        for j in range(i, d):
+            out[j][:3] = path[j + 1][:3]
+        return out
+
+    def unwound_sum(path, i):
+        d = len(path) - 1
+        of, zf = path[i][2], path[i][1]
+        nxt = path[d][3]
+        total = 0.0
+        for j in range(d - 1, -1, -1):
+            if of != 0:
+                tmp = nxt * (d + 1) / ((j + 1) * of)
+                total += tmp
+                nxt = path[j][3] - tmp * zf * (d - j) / (d + 1)
+            elif zf != 0:
+                total += (path[j][3] / zf) * (d + 1) / (d - j)
+        return total
+
+    def rec(node, path, pzf, pof, pfi):
+        path = extend(path, pzf, pof, pfi)
+        l = left[node]
+        if l < 0:
+            for i in range(1, len(path)):
+                w = unwound_sum(path, i)
+                phi[path[i][0]] += w * (path[i][2] - path[i][1]) * value[node]
+            return
+        r = right[node]
+        split = int(feat[node])
+        fv = xr[split]
+        hot = (l if defl[node] else r) if math.isnan(float(fv)) else (l if fv < thresh[node] else r)
+        cold = r if hot == l else l
+        c = cover[node] if cover[node] > 0 else 1.0
+        izf, iof = 1.0, 1.0
+        idx = next((i for i, p in enumerate(path) if p[0] == split), None)
+        if idx is not None:
+            izf, iof = path[idx][1], path[idx][2]
+            path = unwind(path, idx)
+        rec(hot, path, cover[hot] / c * izf, iof, split)
+        rec(cold, path, cover[cold] / c * izf, 0.0, split)
+
+    rec(root, [], 1.0, 1.0, -1)
+
+
+def tree_shap(flat, X, k, t_begin=0, t_end=None):
+    """Exact TreeSHAP contributions: (n, k, f+1) float64 (bias column NOT
+    filled — the booster adds expected values + base margin)."""
+    if t_end is None:
+        t_end = flat["n_trees"]
+    X = X.cpu().contiguous()
+    n, f = X.shape
+    phi = torch.zeros((n, k, f + 1), dtype=torch.float64)
+    if t_end <= t_begin:
+        return phi
+    try:
+        from . import _smxgb_hip as K
+
+        i32 = _flat_i32(flat)
+        K.tree_shap_cpu(
+            X, i32["left"], i32["right"], i32["feature"],
+            flat["threshold"].contiguous(), i32["default_left"],
+            flat["value"].contiguous(), flat["cover"].contiguous(),
+            i32["tree_root"], i32["tree_cls"], t_begin, t_end, phi, k,
+            int(flat.get("max_depth", 32)),
+        )
+        return phi
+    except ImportError:
+        pass
+    Xn = X.numpy()
+    roots = flat["tree_root"].numpy()
+    cls = flat["tree_cls"].numpy()
+    phi_np = phi.numpy()
+    for row in range(n):
+        for t in range(t_begin, t_end):
+            _py_tree_shap_one(flat, Xn[row], phi_np[row, cls[t]], int(roots[t]))
+    return phi
+
+
+def tree_expected_values(flat, k, t_begin=0, t_end=None):
+    """Per-class expected margin of trees [t_begin, t_end): sum over trees
+    of the cover-weighted leaf mean (the TreeSHAP bias term)."""
+    if t_end is None:
+        t_end = flat["n_trees"]
+    left = flat["left"]
+    value = flat["value"].to(torch.float64)
+    cover = flat["cover"].to(torch.float64)
+    roots = flat["tree_root"]
+    cls = flat["tree_cls"]
+    n_nodes = left.shape[0]
+    bounds = torch.cat([roots, torch.tensor([n_nodes], dtype=roots.dtype)])
+    ev = torch.zeros(k, dtype=torch.float64)
+    leaf = left < 0
+    for t in range(t_begin, t_end):
+        s, e = int(bounds[t]), int(bounds[t + 1])
+        lmask = leaf[s:e]
+        croot = float(cover[s])
+        if croot <= 0:
+            continue
+        ev[cls[t]] += float((value[s:e][lmask] * cover[s:e][lmask]).sum()) / croot
+    return ev
+
+
+def pred_leaf(flat, X, t_begin=0, t_end=None):
+    """(n, T) int32 tree-local leaf indices (xgboost pred_leaf=True)."""
+    if t_end is None:
+        t_end = flat["n_trees"]
+    X = X.cpu().contiguous()
+    n = X.shape[0]
+    T = t_end - t_begin
+    out = torch.zeros((n, T), dtype=torch.int32)
+    if T <= 0:
+        return out
+    try:
+        from . import _smxgb_hip as K
+
+        i32 = _flat_i32(flat)
+        K.pred_leaf_cpu(
+            X, i32["left"], i32["right"], i32["feature"],
+            flat["threshold"].contiguous(), i32["default_left"],
+            i32["tree_root"], t_begin, t_end, out,
+        )
+        return out
+    except ImportError:
+        pass
+    # vectorized torch fallback: iterate depth levels for all (row, tree)
+    left = flat["left"]
+    right = flat["right"]
+    feat = flat["feature"]
+    thresh = flat["threshold"]
+    defl = flat["default_left"]
+    roots = flat["tree_root"][t_begin:t_end]
+    node = roots.unsqueeze(0).expand(n, T).contiguous()
+    active = left[node] >= 0
+    while bool(active.any()):
+        nid = node[active]
+        fv = X[active.any(dim=1).nonzero(as_tuple=True)[0], :] if False else None  # noqa
+        rows = active.nonzero(as_tuple=True)[0]
+        fvals = X[rows, feat[nid]]
+        goleft = torch.where(torch.isnan(fvals), defl[nid].to(torch.bool), fvals < thresh[nid])
+        node[active] = torch.where(goleft, left[nid], right[nid]).to(node.dtype)
+        active = left[node] >= 0
+    return (node - roots.unsqueeze(0)).to(torch.int32)
