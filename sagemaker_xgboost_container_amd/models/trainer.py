@@ -121,6 +121,23 @@ def train(
         )
     )
 
+    # tree_method routing: this framework implements ONE updater — the
+    # quantized-histogram grower (the reference's hist/gpu_hist). `exact`
+    # and `approx` are accepted for configuration compatibility (reference
+    # hyperparameter_validation.py:22-25) but deliberately mapped onto
+    # hist; with exact cut finding (ops/quantize.py) and max_bin bins the
+    # models are near-identical. Warn loudly instead of diverging silently.
+    tree_method = params.get("tree_method", "auto")
+    if tree_method in ("exact", "approx"):
+        logger.warning(
+            "tree_method='%s' is not implemented natively; training with the "
+            "histogram updater (tree_method=hist) instead. Results are "
+            "near-identical for continuous features; increase max_bin "
+            "(currently %s) to tighten the approximation.",
+            tree_method,
+            params.get("max_bin", 256),
+        )
+
     device = _resolve_device(params)
     seed = int(params.get("seed", 0) or 0)
     generator = torch.Generator(device=device)

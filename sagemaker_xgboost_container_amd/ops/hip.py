@@ -691,7 +691,20 @@ class DeviceGrower:
                 _ROWS_PER_BLOCK, _GROW_HIST_GRID, self.lds_words, self.hist_block,
             )
             if comm is not None:
-                comm.allreduce_(acc_d)
+                if k == 1:
+                    comm.allreduce_(acc_d)
+                else:
+                    # Compacted allreduce: of each sibling pair exactly one
+                    # slot was built from rows (the other is all zeros —
+                    # make_level's subtraction trick), so folding pairs
+                    # halves the message without a gather kernel. The sum
+                    # lands back in BOTH child slots; convert_level only
+                    # reads the built one. int64 adds wrap mod 2^64,
+                    # matching the kernel's unsigned fixed-point.
+                    compact = acc_d[0::2] + acc_d[1::2]
+                    comm.allreduce_(compact)
+                    acc_d[0::2] = compact
+                    acc_d[1::2] = compact
             hist_d = self.hist_f32[base : base + k]
             _K.grow_convert_level(acc_d, hist_d, nodes_d, k, self.slots2, scale)
             if d > 0:

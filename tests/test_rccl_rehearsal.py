@@ -74,10 +74,11 @@ def _single_rank_worker(port, q):
         rng = np.random.default_rng(3)
         X = rng.normal(size=(100_000, 10)).astype(np.float32)
         y = (X[:, 0] - 0.3 * X[:, 2] > 0).astype(np.float32)
+        params = {"objective": "binary:logistic", "max_depth": 6, "eta": 0.4,
+                  "device": "cuda", "eval_metric": ["logloss"]}
         res = {}
-        trainer.train(
-            {"objective": "binary:logistic", "max_depth": 6, "eta": 0.4, "device": "cuda",
-             "eval_metric": ["logloss"]},
+        bst = trainer.train(
+            params,
             DMatrix(X, label=y),
             num_boost_round=5,
             evals=[(DMatrix(X, label=y), "train")],
@@ -85,7 +86,18 @@ def _single_rank_worker(port, q):
             verbose_eval=False,
             comm=comm,
         )
-        q.put(("ok", res["train"]["logloss"][-1]))
+        # a 1-rank allreduce is the identity: the comm path (per-level
+        # compacted histogram allreduce) must grow EXACTLY the trees the
+        # communicator-free device path grows
+        bst_ref = trainer.train(
+            params, DMatrix(X, label=y), num_boost_round=5, verbose_eval=False,
+        )
+        same_trees = json.dumps(
+            bst.save_json()["learner"]["gradient_booster"]["model"]["trees"], sort_keys=True
+        ) == json.dumps(
+            bst_ref.save_json()["learner"]["gradient_booster"]["model"]["trees"], sort_keys=True
+        )
+        q.put(("ok", res["train"]["logloss"][-1], same_trees))
         dist.barrier()
         dist.destroy_process_group()
     except Exception as e:  # noqa: BLE001 - relayed to the test process
@@ -98,10 +110,11 @@ def test_nccl_single_rank_full_path():
     q = ctx.Queue()
     p = ctx.Process(target=_single_rank_worker, args=(port, q))
     p.start()
-    status, payload = q.get(timeout=420)
+    result = q.get(timeout=420)
     p.join(timeout=120)
-    assert status == "ok", f"nccl 1-rank path failed: {payload}"
-    assert payload < 0.5  # loss decreased through the nccl-comm grower
+    assert result[0] == "ok", f"nccl 1-rank path failed: {result[1]}"
+    assert result[1] < 0.5  # loss decreased through the nccl-comm grower
+    assert result[2], "comm-path trees differ from the communicator-free path"
     assert p.exitcode == 0
 
 

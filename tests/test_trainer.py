@@ -346,3 +346,28 @@ class TestObjectiveZoo:
         bst = trainer.train(params, dtrain, num_boost_round=3, verbose_eval=False)
         pred = bst.predict(dtrain)
         assert np.isfinite(pred).all()
+
+
+def test_exact_tree_method_warns_and_trains(caplog):
+    """`exact`/`approx` are deliberately mapped to hist with a loud warning
+    (reference accepts them via the HP schema; silent rerouting was a
+    round-1 finding)."""
+    import logging as _logging
+
+    import numpy as np
+
+    from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+    from sagemaker_xgboost_container_amd.models import trainer as _trainer
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(300, 4)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    with caplog.at_level(_logging.WARNING):
+        bst = _trainer.train(
+            {"objective": "binary:logistic", "tree_method": "exact", "max_depth": 3},
+            DMatrix(X, label=y),
+            num_boost_round=2,
+            verbose_eval=False,
+        )
+    assert any("tree_method='exact'" in r.message for r in caplog.records)
+    assert len(bst.trees) == 2
