@@ -87,7 +87,7 @@ class HistGrower:
         self.qm = qm
         self.p = GrowParams(params)
         self.comm = comm  # optional: allreduce_(tensor) / allreduce_max_(tensor) / rank
-        self.backend = ops.backend_for(qm.device)
+        self.backend = ops.backend_for_qm(qm)
         self.device = qm.device
         self.generator = generator
         self.state = None

@@ -42,6 +42,34 @@ _TRAIN_PARAM_KEYS = {
 }
 
 
+def _sparse_path_ok(dtrain, params, device, comm):
+    """Engage the CSR training backend (ops/sparse_ref.py)?
+
+    Wide sparse data (libsvm channels, reference data_utils.py:361) never
+    densifies; small/narrow data keeps the dense fast path. Forced with
+    SMXGB_SPARSE=1 / disabled with SMXGB_SPARSE=0. The MI355X device path
+    densifies into HBM (288 GB), so this is the CPU-host path only.
+    """
+    import os as _os
+
+    flag = _os.environ.get("SMXGB_SPARSE")
+    if flag == "0":
+        return False
+    if not getattr(dtrain, "is_sparse", False):
+        return False
+    if device.type != "cpu" or comm is not None:
+        return False
+    if params.get("booster", "gbtree") != "gbtree":
+        return False
+    if params.get("process_type") == "update":
+        return False
+    if flag == "1":
+        return True
+    n, f = dtrain.num_row(), dtrain.num_col()
+    nnz = dtrain.csr().nnz
+    return f >= 64 and nnz <= 0.5 * n * f
+
+
 def _resolve_device(params):
     device = params.get("device")
     if device:
@@ -56,12 +84,15 @@ class _EvalSet:
 
     When the eval set IS the training DMatrix (the standard watchlist), it
     aliases the training tensors and margins — no copy, no per-round
-    traversal (`is_train`).
+    traversal (`is_train`). A wide sparse eval set stays CSR (`self.csr`)
+    and is traversed in bounded-memory chunks instead of densified.
     """
 
     def __init__(self, dmatrix, name, device, n_outputs, base_margin_value,
-                 train_dmatrix=None, train_X=None, train_y=None, train_w=None):
+                 train_dmatrix=None, train_X=None, train_y=None, train_w=None,
+                 keep_sparse=False):
         self.name = name
+        self.csr = None
         self.is_train = dmatrix is train_dmatrix
         if self.is_train:
             self.X = train_X
@@ -70,7 +101,11 @@ class _EvalSet:
             self.dmatrix = dmatrix
             self.margin = None  # provided by the trainer each round
             return
-        self.X = torch.as_tensor(dmatrix.to_dense(), dtype=torch.float32, device=device)
+        if keep_sparse and getattr(dmatrix, "is_sparse", False):
+            self.X = None
+            self.csr = dmatrix.csr()
+        else:
+            self.X = torch.as_tensor(dmatrix.to_dense(), dtype=torch.float32, device=device)
         self.y = torch.as_tensor(dmatrix.get_label(), dtype=torch.float32, device=device)
         w = dmatrix.get_weight()
         self.w = (
@@ -79,15 +114,24 @@ class _EvalSet:
             else None
         )
         self.dmatrix = dmatrix
+        n_rows = dmatrix.num_row()
         user_margin = dmatrix.get_base_margin()
         if user_margin is not None:
             self.margin = torch.as_tensor(user_margin, dtype=torch.float32, device=device).reshape(
-                self.X.shape[0], n_outputs
+                n_rows, n_outputs
             )
         else:
             self.margin = torch.full(
-                (self.X.shape[0], n_outputs), float(base_margin_value), dtype=torch.float32, device=device
+                (n_rows, n_outputs), float(base_margin_value), dtype=torch.float32, device=device
             )
+
+    def tree_margin(self, backend, tree):
+        """One tree's margin contribution over this eval set's rows."""
+        if self.X is None:
+            from ..ops.sparse_ref import predict_tree_csr
+
+            return predict_tree_csr(tree, self.csr)
+        return backend.predict_tree(tree, self.X)
 
 
 def train(
@@ -163,7 +207,17 @@ def train(
     booster.num_features = dtrain.num_col()
 
     # -- resident training state ------------------------------------------
-    X = torch.as_tensor(dtrain.to_dense(), dtype=torch.float32, device=device)
+    sparse_mode = _sparse_path_ok(dtrain, params, device, comm)
+    if sparse_mode:
+        Xcsr = dtrain.csr()
+        X = None
+        logger.info(
+            "Sparse training path engaged: %dx%d CSR, %d nnz (density %.4f) — no densification",
+            Xcsr.shape[0], Xcsr.shape[1], Xcsr.nnz, Xcsr.nnz / max(1, Xcsr.shape[0] * Xcsr.shape[1]),
+        )
+    else:
+        Xcsr = None
+        X = torch.as_tensor(dtrain.to_dense(), dtype=torch.float32, device=device)
     y = torch.as_tensor(dtrain.get_label(), dtype=torch.float32, device=device)
     w_raw = dtrain.get_weight()
     weight = (
@@ -177,7 +231,15 @@ def train(
     max_bin = int(params.get("max_bin", 256))
     qm = None
     if params.get("booster", "gbtree") != "gblinear":
-        qm = quantize(X, max_bin=max_bin, sample_weight=weight, comm=comm)
+        if sparse_mode:
+            from ..ops.sparse_ref import quantize_sparse
+
+            qm = quantize_sparse(
+                Xcsr, max_bin=max_bin,
+                sample_weight=weight.cpu().numpy() if weight is not None else None,
+            )
+        else:
+            qm = quantize(X, max_bin=max_bin, sample_weight=weight, comm=comm)
     backend = backend_for(device)
 
     base_margin_value = objective.base_margin(booster.base_score)
@@ -189,11 +251,17 @@ def train(
         margin = torch.full((n, n_outputs), float(base_margin_value), dtype=torch.float32, device=device)
         # warm start: accumulate existing trees' contributions
         for t_idx, tree in enumerate(booster.trees):
-            margin[:, booster.tree_info[t_idx]] += backend.predict_tree(tree, X)
+            if sparse_mode:
+                from ..ops.sparse_ref import predict_tree_csr
+
+                margin[:, booster.tree_info[t_idx]] += predict_tree_csr(tree, Xcsr)
+            else:
+                margin[:, booster.tree_info[t_idx]] += backend.predict_tree(tree, X)
 
     eval_sets = [
         _EvalSet(dm, name, device, n_outputs, base_margin_value,
-                 train_dmatrix=dtrain, train_X=X, train_y=y, train_w=weight)
+                 train_dmatrix=dtrain, train_X=X, train_y=y, train_w=weight,
+                 keep_sparse=sparse_mode)
         for dm, name in (evals or [])
         if dm is not None
     ]
@@ -201,7 +269,7 @@ def train(
     for es in eval_sets:
         if not es.is_train and es.dmatrix.get_base_margin() is None:
             for t_idx, tree in enumerate(booster.trees):
-                es.margin[:, booster.tree_info[t_idx]] += backend.predict_tree(tree, es.X)
+                es.margin[:, booster.tree_info[t_idx]] += es.tree_margin(backend, tree)
 
     eval_metric = params.get("eval_metric")
     if eval_metric is None:
@@ -343,7 +411,7 @@ def train(
                         round_info.append(cls)
                         for es in eval_sets:
                             if not es.is_train:
-                                es.margin[:, cls] += backend.predict_tree(tree, es.X)
+                                es.margin[:, cls] += es.tree_margin(backend, tree)
                     continue
                 for cls in range(n_outputs):
                     gh_cls = gh if n_outputs == 1 else gh[:, cls, :].contiguous()
@@ -355,7 +423,7 @@ def train(
                     round_info.append(cls)
                     for es in eval_sets:
                         if not es.is_train:
-                            es.margin[:, cls] += backend.predict_tree(tree, es.X) * new_tree_scale
+                            es.margin[:, cls] += es.tree_margin(backend, tree) * new_tree_scale
             # rescale dropped trees and restore their (scaled) contribution
             for t in dropped:
                 booster.weight_drop[t] *= dropped_factor

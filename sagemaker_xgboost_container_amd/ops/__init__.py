@@ -28,3 +28,13 @@ def backend_for(device):
     from . import torch_ref
 
     return torch_ref
+
+
+def backend_for_qm(qm):
+    """Backend for a quantized matrix: the sparse backend for
+    SparseQuantizedMatrix, else the device backend."""
+    from . import sparse_ref
+
+    if isinstance(qm, sparse_ref.SparseQuantizedMatrix):
+        return sparse_ref
+    return backend_for(qm.device)

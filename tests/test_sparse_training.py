@@ -1,0 +1,186 @@
+"""Sparse (CSR) training-path tests: parity with the dense path and the
+bounded-memory guarantee for wide libsvm-style data (reference behavior:
+xgb.DMatrix keeps CSR end-to-end, data_utils.py:361)."""
+import os
+
+import numpy as np
+import pytest
+import scipy.sparse as sp
+
+from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+from sagemaker_xgboost_container_amd.models import trainer
+
+
+def _rand_csr(n, f, density, seed=0, labels_from=0):
+    rng = np.random.default_rng(seed)
+    nnz = int(n * f * density)
+    rows = rng.integers(0, n, nnz)
+    cols = rng.integers(0, f, nnz)
+    vals = rng.normal(size=nnz).astype(np.float32)
+    csr = sp.csr_matrix((vals, (rows, cols)), shape=(n, f), dtype=np.float32)
+    csr.sum_duplicates()
+    col = np.asarray(csr[:, labels_from].todense()).ravel()
+    y = (col > 0).astype(np.float32)
+    return csr, y
+
+
+def _trees_json(bst):
+    import json
+
+    return json.dumps(
+        bst.save_json()["learner"]["gradient_booster"]["model"]["trees"], sort_keys=True
+    )
+
+
+class TestSparseDenseParity:
+    def _train_both(self, csr, y, params, rounds=4, evals=False, weight=None):
+        import copy
+
+        results = {}
+        out = {}
+        for mode in ("1", "0"):  # sparse, dense
+            os.environ["SMXGB_SPARSE"] = mode
+            try:
+                res = {}
+                dm = DMatrix(csr.copy(), label=y, weight=weight)
+                kwargs = {}
+                if evals:
+                    kwargs["evals"] = [(dm, "train"), (DMatrix(csr.copy(), label=y), "validation")]
+                bst = trainer.train(
+                    copy.deepcopy(params), dm, num_boost_round=rounds,
+                    evals_result=res, verbose_eval=False, **kwargs,
+                )
+                out[mode] = bst
+                results[mode] = res
+            finally:
+                os.environ.pop("SMXGB_SPARSE", None)
+        return out, results
+
+    def test_identical_trees_binary(self):
+        csr, y = _rand_csr(2000, 80, 0.3, seed=1)
+        out, _ = self._train_both(
+            csr, y, {"objective": "binary:logistic", "max_depth": 4, "eta": 0.4, "device": "cpu"}
+        )
+        assert _trees_json(out["1"]) == _trees_json(out["0"])
+
+    def test_identical_trees_with_eval_history(self):
+        csr, y = _rand_csr(1500, 100, 0.2, seed=2)
+        out, results = self._train_both(
+            csr, y,
+            {"objective": "binary:logistic", "max_depth": 3, "eta": 0.3, "device": "cpu",
+             "eval_metric": ["logloss", "auc"]},
+            evals=True,
+        )
+        assert _trees_json(out["1"]) == _trees_json(out["0"])
+        for ds in ("train", "validation"):
+            for m in ("logloss", "auc"):
+                np.testing.assert_allclose(
+                    results["1"][ds][m], results["0"][ds][m], rtol=1e-6, atol=1e-7
+                )
+
+    def test_identical_trees_weighted_regression(self):
+        csr, y = _rand_csr(1200, 64, 0.25, seed=3)
+        w = np.random.default_rng(4).uniform(0.5, 2.0, 1200).astype(np.float32)
+        out, _ = self._train_both(
+            csr, y.astype(np.float32),
+            {"objective": "reg:squarederror", "max_depth": 4, "eta": 0.3, "device": "cpu"},
+            weight=w,
+        )
+        assert _trees_json(out["1"]) == _trees_json(out["0"])
+
+    def test_explicit_zero_vs_absent(self):
+        # absent entries are missing (default direction); stored zeros are
+        # the value 0 — the two must train identically to the dense path
+        # where absent is NaN and stored zero is 0.0
+        rng = np.random.default_rng(5)
+        n, f = 800, 70
+        csr, y = _rand_csr(n, f, 0.2, seed=5)
+        # inject explicit zeros at random positions
+        zr = rng.integers(0, n, 500)
+        zc = rng.integers(0, f, 500)
+        zeros = sp.csr_matrix((np.zeros(500, np.float32), (zr, zc)), shape=(n, f))
+        merged = csr + zeros  # keeps explicit zeros where csr had none
+        out, _ = self._train_both(
+            merged, y, {"objective": "binary:logistic", "max_depth": 3, "device": "cpu"}
+        )
+        assert _trees_json(out["1"]) == _trees_json(out["0"])
+
+    def test_subsample_colsample_run_on_sparse(self):
+        csr, y = _rand_csr(2000, 80, 0.15, seed=6)
+        os.environ["SMXGB_SPARSE"] = "1"
+        try:
+            bst = trainer.train(
+                {"objective": "binary:logistic", "max_depth": 4, "subsample": 0.7,
+                 "colsample_bytree": 0.6, "seed": 9, "device": "cpu"},
+                DMatrix(csr, label=y),
+                num_boost_round=3,
+                verbose_eval=False,
+            )
+        finally:
+            os.environ.pop("SMXGB_SPARSE", None)
+        assert len(bst.trees) == 3
+        p = bst.predict(np.zeros((2, 80), dtype=np.float32))
+        assert p.shape == (2,)
+
+    def test_auto_engages_on_wide_sparse_only(self, caplog):
+        import logging as _logging
+
+        # narrow data (f < 64) keeps the dense path even when CSR
+        csr, y = _rand_csr(500, 20, 0.3, seed=7)
+        with caplog.at_level(_logging.INFO):
+            trainer.train(
+                {"objective": "binary:logistic", "max_depth": 3, "device": "cpu"},
+                DMatrix(csr, label=y), num_boost_round=1, verbose_eval=False,
+            )
+        assert not any("Sparse training path engaged" in r.message for r in caplog.records)
+        caplog.clear()
+        csr, y = _rand_csr(500, 128, 0.05, seed=8)
+        with caplog.at_level(_logging.INFO):
+            trainer.train(
+                {"objective": "binary:logistic", "max_depth": 3, "device": "cpu"},
+                DMatrix(csr, label=y), num_boost_round=1, verbose_eval=False,
+            )
+        assert any("Sparse training path engaged" in r.message for r in caplog.records)
+
+
+class TestWideSparseMemory:
+    def test_500k_x_20k_trains_in_bounded_memory(self):
+        """VERDICT round-1 item 7: 500k x 20k at 0.1% nnz must train within
+        a few GB (dense would be ~37 GB float + 10 GB bins)."""
+        import resource
+
+        n, f, density = 500_000, 20_000, 0.001
+        csr, y = _rand_csr(n, f, density, seed=11)
+        rss_before = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss  # KiB
+        bst = trainer.train(
+            {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3, "device": "cpu"},
+            DMatrix(csr, label=y),
+            num_boost_round=3,
+            verbose_eval=False,
+        )
+        rss_after = resource.getrusage(resource.RUSAGE_SELF).ru_maxrss
+        grew_gb = (rss_after - rss_before) / 1024 / 1024
+        assert len(bst.trees) == 3
+        assert bst.trees[0].num_nodes > 1  # it actually split
+        assert grew_gb < 4.0, f"sparse training grew RSS by {grew_gb:.2f} GB"
+
+    def test_wide_libsvm_file_end_to_end(self, tmp_path):
+        # a wide libsvm channel parses to CSR and trains sparse
+        rng = np.random.default_rng(12)
+        lines = []
+        for _ in range(400):
+            label = int(rng.random() > 0.5)
+            idx = sorted(rng.choice(5000, size=8, replace=False))
+            feats = " ".join(f"{j}:{rng.normal():.4f}" for j in idx)
+            lines.append(f"{label} {feats}")
+        # guarantee the max feature index appears so num_col is stable
+        lines.append("1 4999:1.5")
+        path = tmp_path / "train.libsvm"
+        path.write_text("\n".join(lines) + "\n")
+        dm = DMatrix(f"{path}?format=libsvm")
+        assert dm.is_sparse
+        bst = trainer.train(
+            {"objective": "binary:logistic", "max_depth": 3, "device": "cpu"},
+            dm, num_boost_round=2, verbose_eval=False,
+        )
+        assert len(bst.trees) == 2
