@@ -277,32 +277,59 @@ class HistGrower:
             and p.colsample_bynode >= 1.0
         )
 
+    def _class_stream(self, slot):
+        """One HIP stream per round-tree slot: a multiclass round's k small
+        per-tree kernels (Covertype-shape trees underfill 256 CUs on their
+        own) execute CONCURRENTLY across class streams instead of
+        back-to-back on one stream."""
+        streams = getattr(self, "_streams", None)
+        if streams is None:
+            streams = {}
+            self._streams = streams
+        s = streams.get(slot)
+        if s is None:
+            s = torch.cuda.Stream(device=self.device)
+            streams[slot] = s
+        return s
+
     def grow_async(self, gh, slot=0):
         """Enqueue one independent tree of a round without waiting.
 
-        Requires device_async_ok(). Each `slot` uses its own compact buffers
-        and heap, so several trees can be in flight on the stream at once —
-        the GPU never idles while the host builds the previous class's tree.
+        Requires device_async_ok(). Each `slot` uses its own compact
+        buffers, heap AND HIP stream, so several trees of one round
+        (multiclass / bagging) run concurrently on the GPU.
         Returns an opaque handle for grow_finish().
         """
         p = self.p
         qm = self.qm
         n = qm.num_row
-        if p.subsample < 1.0:
-            keep = torch.rand(n, device=self.device, generator=self.generator) < p.subsample
-            rows = keep.nonzero(as_tuple=True)[0].to(torch.int32)
-        else:
-            rows = None
-        state = self.backend.make_tree_state(qm, gh, rows, slot=slot)
-        self.state = state
-        scale = self.backend.compute_scale(gh, comm=None)
-        tree_mask = self._sample_features(p.colsample_bytree, None)
-        dg = self._device_grower_for(slot)
-        dg.state = state
-        if tree_mask is not None:
-            dg.mask = tree_mask.to(torch.uint8).contiguous()
-        ev = dg.grow_enqueue(scale, (p.reg_lambda, p.reg_alpha, p.gamma, p.min_child_weight))
-        return {"dg": dg, "ev": ev, "state": state, "cap": state.cap}
+        # per-qm shared caches must exist on the DEFAULT stream before any
+        # class stream reads them (class streams only order against the
+        # default stream, not each other)
+        if not getattr(qm, "_async_warm", False):
+            if getattr(qm, "_arange_cache", None) is None or qm._arange_cache.numel() != n:
+                qm._arange_cache = torch.arange(n, dtype=torch.int32, device=self.device)
+            if not hasattr(qm, "_nbins_i32"):
+                qm._nbins_i32 = qm.nbins.to(torch.int32).contiguous()
+            qm._async_warm = True
+        stream = self._class_stream(slot)
+        stream.wait_stream(torch.cuda.current_stream(self.device))
+        with torch.cuda.stream(stream):
+            if p.subsample < 1.0:
+                keep = torch.rand(n, device=self.device, generator=self.generator) < p.subsample
+                rows = keep.nonzero(as_tuple=True)[0].to(torch.int32)
+            else:
+                rows = None
+            state = self.backend.make_tree_state(qm, gh, rows, slot=slot)
+            self.state = state
+            scale = self.backend.compute_scale(gh, comm=None)
+            tree_mask = self._sample_features(p.colsample_bytree, None)
+            dg = self._device_grower_for(slot)
+            dg.state = state
+            if tree_mask is not None:
+                dg.mask = tree_mask.to(torch.uint8).contiguous()
+            ev = dg.grow_enqueue(scale, (p.reg_lambda, p.reg_alpha, p.gamma, p.min_child_weight))
+        return {"dg": dg, "ev": ev, "state": state, "cap": state.cap, "stream": stream}
 
     def grow_finish(self, handle):
         """Wait for a grow_async tree and build its host Tree."""
