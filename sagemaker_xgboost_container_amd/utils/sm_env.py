@@ -39,7 +39,12 @@ def _read_json(path, default=None):
 class TrainingEnv:
     """The SageMaker training filesystem/env contract."""
 
-    def __init__(self, base_path=BASE_PATH):
+    def __init__(self, base_path=None):
+        # SAGEMAKER_BASE_DIR relocates the whole /opt/ml tree — the
+        # container-boundary test tier fabricates one per test run, same
+        # trick as the reference's local_mode harness (local_mode.py:330-396)
+        if base_path is None:
+            base_path = os.environ.get("SAGEMAKER_BASE_DIR", BASE_PATH)
         self.base_path = base_path
         input_config = os.path.join(base_path, "input", "config")
         self.hyperparameters_file = os.environ.get(
