@@ -57,7 +57,7 @@ def _sparse_path_ok(dtrain, params, device, comm):
         return False
     if not getattr(dtrain, "is_sparse", False):
         return False
-    if device.type != "cpu" or comm is not None:
+    if device.type != "cpu":
         return False
     if params.get("booster", "gbtree") != "gbtree":
         return False
@@ -237,6 +237,7 @@ def train(
             qm = quantize_sparse(
                 Xcsr, max_bin=max_bin,
                 sample_weight=weight.cpu().numpy() if weight is not None else None,
+                comm=comm,
             )
         else:
             qm = quantize(X, max_bin=max_bin, sample_weight=weight, comm=comm)

@@ -62,30 +62,98 @@ class SparseQuantizedMatrix:
         return self.stride - 1
 
 
-def quantize_sparse(csr, max_bin=256, sample_weight=None):
+def _merged_cuts_distributed(csc, f, max_bin, w, comm, n_candidates=None):
+    """Globally consistent per-feature cuts across ranks for sparse data:
+    each rank summarizes each feature's PRESENT values as n_candidates
+    quantile positions carrying local_mass/n_candidates weight each; the
+    summaries allgather (one fused collective) and the merged weighted
+    multiset is re-quantiled — identical on every rank by construction
+    (same scheme as ops/quantize.make_cuts_distributed for dense)."""
+    ncand = n_candidates or min(4 * max_bin, 4096)
+    cand = torch.full((f, ncand), float("nan"))
+    mass = torch.zeros(f)
+    for j in range(f):
+        s, e = int(csc.indptr[j]), int(csc.indptr[j + 1])
+        m = e - s
+        if m == 0:
+            continue
+        vals = torch.from_numpy(np.ascontiguousarray(csc.data[s:e], dtype=np.float32))
+        vals, order = torch.sort(vals)
+        if w is not None:
+            wj = torch.from_numpy(w[csc.indices[s:e]])[order]
+            cw = torch.cumsum(wj, 0)
+            total = cw[-1]
+            targets = (torch.arange(ncand, dtype=torch.float32) + 0.5) / ncand * total
+            pos = torch.searchsorted(cw, targets).clamp_(0, m - 1)
+            cand[j] = vals[pos]
+            mass[j] = float(total)
+        else:
+            pos = ((torch.arange(ncand, dtype=torch.float64) + 0.5) / ncand * m).long()
+            cand[j] = vals[pos.clamp_(0, m - 1)]
+            mass[j] = float(m)
+
+    world = comm.world_size
+    all_cand = torch.zeros((world, f, ncand))
+    all_mass = torch.zeros((world, f))
+    all_cand[comm.rank] = cand
+    all_mass[comm.rank] = mass
+    comm.allreduce_(all_cand)
+    comm.allreduce_(all_mass)
+
+    cut_list = []
+    for j in range(f):
+        values = all_cand[:, j, :].reshape(-1)
+        weights = (all_mass[:, j] / ncand).reshape(-1, 1).expand(-1, ncand).reshape(-1)
+        keep = (~torch.isnan(values)) & (weights > 0)
+        values, weights = values[keep], weights[keep]
+        if values.numel() == 0:
+            cut_list.append(torch.zeros(0))
+            continue
+        distinct = torch.unique(values)
+        if distinct.numel() <= max_bin:
+            cuts_j = (distinct[:-1] + distinct[1:]) * 0.5 if distinct.numel() > 1 else torch.zeros(0)
+        else:
+            sorted_vals, order = torch.sort(values)
+            cw = torch.cumsum(weights[order], 0)
+            total = cw[-1]
+            targets = torch.linspace(0, 1, max_bin + 1)[1:-1] * total
+            pos = torch.searchsorted(cw, targets).clamp_(0, values.numel() - 1)
+            cuts_j = torch.unique(sorted_vals[pos])
+        cut_list.append(cuts_j)
+    return cut_list
+
+
+def quantize_sparse(csr, max_bin=256, sample_weight=None, comm=None):
     """Quantize a scipy CSR matrix into a SparseQuantizedMatrix.
 
     Cut semantics match ops/quantize.make_cuts on the NaN-densified
     matrix: per-feature exact quantiles over present values, midpoint
-    cuts when <= max_bin distinct values.
+    cuts when <= max_bin distinct values. With a communicator, cuts come
+    from the merged GLOBAL distribution so every rank bins identically.
     """
     csr = csr.tocsr()
     n, f = csr.shape
     csc = csr.tocsc()
     w = None if sample_weight is None else np.asarray(sample_weight, dtype=np.float32)
 
-    cut_list = []
     nbins = np.empty(f, dtype=np.int64)
     csc_bins = np.empty(csc.data.shape[0], dtype=np.int64)
+    merged = None
+    if comm is not None and comm.world_size > 1:
+        merged = _merged_cuts_distributed(csc, f, max_bin, w, comm)
+    cut_list = []
     for j in range(f):
         s, e = int(csc.indptr[j]), int(csc.indptr[j + 1])
         vals = torch.from_numpy(np.ascontiguousarray(csc.data[s:e], dtype=np.float32))
-        wj = None
-        if w is not None and e > s:
-            wj = torch.from_numpy(w[csc.indices[s:e]])
-        from .quantize import _feature_cuts
+        if merged is not None:
+            cuts_j = merged[j]
+        else:
+            wj = None
+            if w is not None and e > s:
+                wj = torch.from_numpy(w[csc.indices[s:e]])
+            from .quantize import _feature_cuts
 
-        cuts_j = _feature_cuts(vals, max_bin, wj)
+            cuts_j = _feature_cuts(vals, max_bin, wj)
         cut_list.append(cuts_j)
         nbins[j] = cuts_j.numel() + 1
         if e > s:

@@ -184,3 +184,69 @@ class TestWideSparseMemory:
             dm, num_boost_round=2, verbose_eval=False,
         )
         assert len(bst.trees) == 2
+
+
+class TestDistributedSparse:
+    @staticmethod
+    def _worker(rank, world, port, q):
+        import datetime
+        import hashlib
+        import json
+
+        import torch.distributed as dist
+
+        from sagemaker_xgboost_container_amd.parallel.comm import Communicator
+
+        try:
+            dist.init_process_group(
+                backend="gloo", init_method=f"tcp://127.0.0.1:{port}",
+                rank=rank, world_size=world, timeout=datetime.timedelta(seconds=120),
+            )
+            comm = Communicator()
+            csr, y = _rand_csr(3000, 128, 0.05, seed=21)
+            sl = np.arange(rank, 3000, world)
+            res = {}
+            os.environ["SMXGB_SPARSE"] = "1"
+            bst = trainer.train(
+                {"objective": "binary:logistic", "max_depth": 4, "eta": 0.4,
+                 "device": "cpu", "eval_metric": ["logloss"]},
+                DMatrix(csr[sl], label=y[sl]),
+                num_boost_round=4,
+                evals=[(DMatrix(csr[sl], label=y[sl]), "train")],
+                evals_result=res,
+                verbose_eval=False,
+                comm=comm,
+            )
+            sig = hashlib.sha256(json.dumps(
+                bst.save_json()["learner"]["gradient_booster"]["model"]["trees"],
+                sort_keys=True).encode()).hexdigest()
+            q.put(("ok", rank, sig, res["train"]["logloss"][-1]))
+            dist.barrier()
+            dist.destroy_process_group()
+        except Exception:  # noqa: BLE001
+            import traceback
+
+            q.put(("error", rank, traceback.format_exc(), None))
+
+    def test_two_rank_sparse_identical_trees(self):
+        import multiprocessing as mp
+        import socket
+
+        s = socket.socket()
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+        s.close()
+        ctx = mp.get_context("spawn")
+        q = ctx.Queue()
+        procs = [ctx.Process(target=self._worker, args=(r, 2, port, q)) for r in range(2)]
+        for p in procs:
+            p.start()
+        results = [q.get(timeout=300) for _ in range(2)]
+        for p in procs:
+            p.join(timeout=60)
+        errors = [r for r in results if r[0] == "error"]
+        assert not errors, "\n".join(str(e[2]) for e in errors)
+        sigs = {r[2] for r in results}
+        assert len(sigs) == 1, "ranks grew different trees on the sparse path"
+        assert results[0][3] == pytest.approx(results[1][3], abs=1e-12)
+        assert results[0][3] < 0.6
