@@ -627,8 +627,11 @@ class DeviceGrower:
         alias the matrix; only gh changes round to round — staged) and no
         per-tree feature mask (its contents change every tree)."""
         st = self.state
+        # default OFF: capture/replay of this sequence currently memory-
+        # faults on gfx950 (gpurun_out/graph_triage.log); opt in with
+        # SMXGB_HIPGRAPH=1 once the faulting node is resolved
         return (
-            _os.environ.get("SMXGB_HIPGRAPH", "1") != "0"
+            _os.environ.get("SMXGB_HIPGRAPH", "0") == "1"
             and st._bins_init is self.qm.bins
             and self.mask.numel() == 0
         )
