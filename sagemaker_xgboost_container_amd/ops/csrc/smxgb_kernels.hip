@@ -1313,6 +1313,18 @@ static void launch_hist_device(int grid, int hist_block, size_t lds_bytes, hipSt
                                unsigned long long* acc, int k, int nfeat, int stride,
                                int n_groups, int feats_per_group, const float* gh_max,
                                int rows_per_block) {
+  // hipGraph kernel nodes validate dynamic LDS against the function's
+  // max-dynamic-shared attribute (plain launches do not): without this
+  // opt-in a captured full-slab launch (>64 KB) reads a truncated LDS
+  // allocation and memory-faults on replay (gfx950 allows 160 KB).
+  static bool lds_attr_set = [] {
+    (void)hipFuncSetAttribute((const void*)(hist_device_kernel<BinT, 512>),
+                              hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    (void)hipFuncSetAttribute((const void*)(hist_device_kernel<BinT, HIST_BLOCK>),
+                              hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    return true;
+  }();
+  (void)lds_attr_set;
   if (hist_block == 512) {
     hipLaunchKernelGGL((hist_device_kernel<BinT, 512>), dim3(grid), dim3(512), lds_bytes, stream,
                        bins_c, gh_c, nodes, hist_prefix, work, acc, k, nfeat, stride, n_groups,
