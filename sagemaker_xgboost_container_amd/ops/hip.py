@@ -655,8 +655,18 @@ class DeviceGrower:
             # paths: the fused-gradient partial sum (when present) is
             # staged, not recomputed, so graph/non-graph grow the same tree
             key = (st.cap, split_params, rootsum is not None)
+            dbg = _os.environ.get("SMXGB_GRAPH_DEBUG") == "1"
+
+            def _d(msg):
+                if dbg:
+                    import sys as _sys
+
+                    print(f"[graphdbg] {msg}", file=_sys.stderr, flush=True)
+                    torch.cuda.synchronize()
+
             if getattr(self, "_graph_key", None) != key:
                 try:
+                    _d(f"build start cap={st.cap}")
                     self._gh_stage = torch.empty(
                         (st.cap, 2), dtype=torch.float32, device=self.nodes.device
                     )
@@ -674,10 +684,12 @@ class DeviceGrower:
                                            rootsum=self._root_stage)
                     torch.cuda.current_stream().wait_stream(side)
                     torch.cuda.synchronize()
+                    _d("warmup done")
                     graph = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(graph):
                         self._enqueue_body(self._gh_stage, self._scale_stage, split_params,
                                            rootsum=self._root_stage)
+                    _d("capture done")
                     self._graph = graph
                     self._graph_key = key
                 except Exception as e:  # capture unsupported -> plain enqueue
@@ -696,6 +708,11 @@ class DeviceGrower:
             if self._root_stage is not None:
                 self._root_stage.copy_(rootsum, non_blocking=True)
             self._graph.replay()
+            if _os.environ.get("SMXGB_GRAPH_DEBUG") == "1":
+                import sys as _sys
+
+                torch.cuda.synchronize()
+                print("[graphdbg] replay ok", file=_sys.stderr, flush=True)
         else:
             self._enqueue_body(st._gh_init, scale, split_params, rootsum=rootsum)
         ev = torch.cuda.Event()
