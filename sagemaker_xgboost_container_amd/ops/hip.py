@@ -583,9 +583,17 @@ class DeviceGrower:
         if not hasattr(qm, "_nbins_i32"):
             qm._nbins_i32 = qm.nbins.to(torch.int32).contiguous()
 
-    def _enqueue_body(self, gh_init, scale, split_params, rootsum=None):
+    def _enqueue_body(self, gh_init, scale, split_params, rootsum=None, alloc_free=False):
         """The tree's full kernel sequence + pinned readback. Pointer-stable
-        given (gh_init, scale) tensors — hipGraph-capturable."""
+        given (gh_init, scale) tensors — hipGraph-capturable.
+
+        alloc_free=True (the captured variant) requires `rootsum` staged by
+        the caller and performs ZERO allocator calls: capture-time
+        allocations route through the graph's private pool, and on this
+        ROCm stack later eager allocations between replays corrupt replayed
+        output (measured: gpurun_out/graph_il2.log — replay #2 returns
+        wrong splits once the host allocates between replays).
+        """
         st = self.state
         qm = self.qm
         f = qm.num_col
@@ -594,10 +602,15 @@ class DeviceGrower:
         reg_lambda, reg_alpha, gamma, mcw = split_params
 
         self.counts.zero_()
-        # full-data (G, H): the fused gradient kernel attaches its one-pass
-        # partial-sum result; recompute only for subsampled/torch-path gh
-        root_gh = rootsum.clone() if rootsum is not None else gh_init.to(torch.float64).sum(0)
-        self.node_gh[0] = root_gh.to(torch.float32)
+        if alloc_free:
+            # f64 -> f32 conversion happens inside copy_ — no temporaries
+            self.node_gh[0].copy_(rootsum)
+        else:
+            # full-data (G, H): the fused gradient kernel attaches its
+            # one-pass partial-sum result; recompute only for
+            # subsampled/torch-path gh
+            root_gh = rootsum.clone() if rootsum is not None else gh_init.to(torch.float64).sum(0)
+            self.node_gh[0] = root_gh.to(torch.float32)
 
         # whole tree enqueued from ONE extension call
         _K.grow_tree_enqueue(
@@ -653,8 +666,11 @@ class DeviceGrower:
         if self._graph_eligible():
             # root (G, H) must stay BIT-identical with the non-graphed
             # paths: the fused-gradient partial sum (when present) is
-            # staged, not recomputed, so graph/non-graph grow the same tree
-            key = (st.cap, split_params, rootsum is not None)
+            # staged; otherwise the same f64 sum runs EAGERLY here (the
+            # captured body must be allocation-free — see _enqueue_body)
+            if rootsum is None:
+                rootsum = st._gh_init.to(torch.float64).sum(0)
+            key = (st.cap, split_params)
             dbg = _os.environ.get("SMXGB_GRAPH_DEBUG") == "1"
 
             def _d(msg):
@@ -671,24 +687,22 @@ class DeviceGrower:
                         (st.cap, 2), dtype=torch.float32, device=self.nodes.device
                     )
                     self._scale_stage = torch.empty_like(scale)
-                    self._root_stage = (
-                        torch.empty(2, dtype=torch.float64, device=self.nodes.device)
-                        if rootsum is not None
-                        else None
+                    self._root_stage = torch.empty(
+                        2, dtype=torch.float64, device=self.nodes.device
                     )
                     # warmup on a side stream (allocator settles), then capture
                     side = torch.cuda.Stream(device=self.nodes.device)
                     side.wait_stream(torch.cuda.current_stream())
                     with torch.cuda.stream(side):
                         self._enqueue_body(self._gh_stage, self._scale_stage, split_params,
-                                           rootsum=self._root_stage)
+                                           rootsum=self._root_stage, alloc_free=True)
                     torch.cuda.current_stream().wait_stream(side)
                     torch.cuda.synchronize()
                     _d("warmup done")
                     graph = torch.cuda.CUDAGraph()
                     with torch.cuda.graph(graph):
                         self._enqueue_body(self._gh_stage, self._scale_stage, split_params,
-                                           rootsum=self._root_stage)
+                                           rootsum=self._root_stage, alloc_free=True)
                     _d("capture done")
                     self._graph = graph
                     self._graph_key = key
@@ -705,8 +719,7 @@ class DeviceGrower:
                     return ev
             self._gh_stage.copy_(st._gh_init, non_blocking=True)
             self._scale_stage.copy_(scale, non_blocking=True)
-            if self._root_stage is not None:
-                self._root_stage.copy_(rootsum, non_blocking=True)
+            self._root_stage.copy_(rootsum, non_blocking=True)
             self._graph.replay()
             if _os.environ.get("SMXGB_GRAPH_DEBUG") == "1":
                 import sys as _sys
