@@ -554,6 +554,15 @@ class DeviceGrower:
         self.state = state
         self.qm = state.qm
         self.D = max_depth
+        # virtual-block granularity adapts to the tree's row count: small
+        # trees (multiclass class-trees, small data) need more blocks in
+        # flight to fill 256 CUs — measured 141 -> 178 r/s on Covertype-
+        # shape at 2048 vs 4096 (profiles/r02_optimization_log.md); the
+        # 12.5M-row flagship keeps 4096 (267 vs 257 at 2048).
+        if _os.environ.get("SMXGB_ROWS_PER_BLOCK"):
+            self.rows_per_block = _ROWS_PER_BLOCK
+        else:
+            self.rows_per_block = 4096 if state.cap >= (1 << 22) else 2048
         qm = state.qm
         device = qm.bins.device
         f = qm.num_col
@@ -622,7 +631,7 @@ class DeviceGrower:
             qm._nbins_i32, self.mask, scale,
             self.D, st.cap, f, stride, self.n_groups, self.feats_per_group,
             self.lds_words, 1 if qm.has_missing else 0, missing_bin,
-            _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB, _GROW_HIST_GRID, _GROW_PART_GRID,
+            self.rows_per_block, _MAX_BLOCKS_PER_JOB, _GROW_HIST_GRID, _GROW_PART_GRID,
             reg_lambda, reg_alpha, gamma, mcw, self.hist_block,
         )
         if not hasattr(self, "_pinned"):
@@ -772,7 +781,7 @@ class DeviceGrower:
 
             if d == 0:
                 _K.grow_make_root(nodes_d, self.hp[0], self.pp[0], self.work[0],
-                                  st.cap, _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB)
+                                  st.cap, self.rows_per_block, _MAX_BLOCKS_PER_JOB)
             else:
                 pk = k >> 1
                 pbase = pk - 1
@@ -780,7 +789,7 @@ class DeviceGrower:
                     self.nodes[pbase : pbase + pk], self.splits[pbase : pbase + pk],
                     self.counts[pbase : pbase + pk], self.node_gh[pbase : pbase + pk],
                     nodes_d, gh_d, self.hp[d], self.pp[d], self.work[d],
-                    pk, _ROWS_PER_BLOCK, _MAX_BLOCKS_PER_JOB,
+                    pk, self.rows_per_block, _MAX_BLOCKS_PER_JOB,
                 )
 
             src_bins, src_gh, src_rows = (
@@ -792,7 +801,7 @@ class DeviceGrower:
             _K.grow_hist_level(
                 src_bins, src_gh, nodes_d, self.hp[d], self.work[d], acc_d,
                 k, f, stride, self.n_groups, self.feats_per_group, scale,
-                _ROWS_PER_BLOCK, _GROW_HIST_GRID, self.lds_words, self.hist_block,
+                self.rows_per_block, _GROW_HIST_GRID, self.lds_words, self.hist_block,
             )
             if comm is not None:
                 if k == 1:
