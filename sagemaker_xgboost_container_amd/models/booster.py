@@ -211,8 +211,17 @@ class Booster:
                     f"feature_names mismatch: model expects {self.num_features} features, "
                     f"got {arr.shape[1]}"
                 )
-        device = "cuda" if torch.cuda.is_available() and self.params.get("predictor") != "cpu_predictor" else "cpu"
-        return torch.as_tensor(arr, dtype=torch.float32, device=device)
+        predictor = self.params.get("predictor")
+        use_gpu = torch.cuda.is_available() and predictor != "cpu_predictor"
+        if use_gpu and predictor != "gpu_predictor":
+            # auto: tiny batches are launch-bound on GPU — the parallel C++
+            # traversal wins below ~150k row-trees (measured crossover,
+            # benchmarks/bench_predict_cpu_gpu.py: 100x500 trees CPU 0.70ms
+            # vs GPU 0.92ms; 1000x500 GPU 1.03ms vs CPU 7.2ms)
+            n_rows = arr.shape[0] if hasattr(arr, "shape") else len(arr)
+            if n_rows * max(1, len(self.trees)) < 150_000:
+                use_gpu = False
+        return torch.as_tensor(arr, dtype=torch.float32, device="cuda" if use_gpu else "cpu")
 
     def _cpu_flat_forest(self):
         """Flat forest on host for the contrib/leaf paths (cached)."""
