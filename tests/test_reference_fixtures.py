@@ -234,3 +234,23 @@ class TestDataFixtures:
         )
         rmse = res["train"]["rmse"]
         assert rmse[-1] < rmse[0]
+
+
+class TestWeightSidecarFixtures:
+    def test_reference_weights_files_validate(self):
+        from sagemaker_xgboost_container_amd.data import data_utils
+
+        for p, fmt in [
+            (f"{REF}/data/csv/train.csv.weights", "csv"),
+            (f"{REF}/data/libsvm/train.libsvm.weights", "libsvm"),
+        ]:
+            data_utils.validate_data_file_path(p, fmt)  # must not raise
+
+    def test_inline_libsvm_weights_parse(self):
+        # <label>:<weight> syntax (reference data_utils.py:155,186)
+        from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+
+        dm = DMatrix(f"{REF}/data/libsvm/train.libsvm.weights?format=libsvm")
+        w = dm.get_weight()
+        assert len(w) == dm.num_row()
+        assert abs(float(w[0]) - 0.2) < 1e-6
