@@ -247,19 +247,26 @@ def _stage_with_depth_warning(dest_path, data_path):
 
 
 def _get_file_mode_files_path(data_path):
-    """Stage input files/dirs into one flat dir of symlinks; return its path."""
+    """Stage input files/dirs into one flat dir of symlinks; return its path.
+
+    The staging dir is per-process: the multi-GPU runner loads channels in
+    EVERY worker process concurrently, and a shared staging dir would have
+    one rank rmtree-ing the symlinks another rank is reading (observed as
+    mkdir/read races in the 2-worker runner test).
+    """
     logging.info("File path %s of input files", data_path)
-    shutil.rmtree(STAGING_DIR, ignore_errors=True)
-    os.mkdir(STAGING_DIR)
+    staging = f"{STAGING_DIR}-{os.getpid()}"
+    shutil.rmtree(staging, ignore_errors=True)
+    os.mkdir(staging)
     if isinstance(data_path, list):
         for path in data_path:
-            _stage_with_depth_warning(STAGING_DIR, path)
+            _stage_with_depth_warning(staging, path)
     else:
         if not os.path.exists(data_path):
             logging.info("File path %s does not exist!", data_path)
             return None
-        _stage_with_depth_warning(STAGING_DIR, data_path)
-    return STAGING_DIR
+        _stage_with_depth_warning(staging, data_path)
+    return staging
 
 
 def _get_pipe_mode_files_path(data_path):

@@ -66,9 +66,13 @@ def _worker(local_rank, num_gpus, sm_hosts, current_host, hyperparameters, train
     master_addr = os.environ.get("MASTER_ADDR") or (hosts[0] if len(hosts) > 1 else "127.0.0.1")
     master_port = os.environ.get("MASTER_PORT", "23456")
 
-    torch.cuda.set_device(local_rank)
+    # nccl(=RCCL) on GPU instances; gloo keeps the same code path testable
+    # (and running) on CPU-only hosts
+    use_cuda = torch.cuda.is_available()
+    if use_cuda:
+        torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
     dist.init_process_group(
-        backend="nccl",
+        backend="nccl" if use_cuda else "gloo",
         init_method=f"tcp://{master_addr}:{master_port}",
         rank=rank,
         world_size=world_size,
@@ -99,7 +103,7 @@ def _worker(local_rank, num_gpus, sm_hosts, current_host, hyperparameters, train
 
     hp = dict(hyperparameters)
     hp["tree_method"] = GPU_TREE_METHOD
-    hp["device"] = f"cuda:{local_rank}"
+    hp["device"] = f"cuda:{local_rank % max(torch.cuda.device_count(), 1)}" if use_cuda else "cpu"
     train_job(
         train_cfg=hp,
         train_dmatrix=shard,

@@ -1,0 +1,120 @@
+"""End-to-end tests for the multi-GPU training runner
+(`distributed_gpu/distributed_gpu_training.run_training_with_rccl`) — the
+replacement for the reference's Dask-GPU stack and the path
+`use_dask_gpu_training=true` maps onto. On CPU hosts the same code runs
+over gloo (2 spawned workers), so the full flow — channel loading, row
+sharding, VALIDATION sharding (every rank must own an eval set or the
+fused metric allreduce deadlocks — advisor r01 high finding), early
+stopping, rank-0 model save — executes here without hardware.
+"""
+import json
+import os
+
+import numpy as np
+import pytest
+
+from sagemaker_xgboost_container_amd.distributed_gpu.distributed_gpu_training import (
+    run_training_with_rccl,
+    validate_gpu_train_configuration,
+)
+
+
+@pytest.fixture
+def channels(tmp_path):
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(3000, 6)).astype(np.float32)
+    y = (X[:, 0] + 0.5 * X[:, 1] > 0).astype(np.float32)
+    train_dir = tmp_path / "train"
+    val_dir = tmp_path / "validation"
+    train_dir.mkdir()
+    val_dir.mkdir()
+    np.savetxt(train_dir / "part0.csv", np.column_stack([y[:2500], X[:2500]]), delimiter=",", fmt="%.5f")
+    np.savetxt(val_dir / "part0.csv", np.column_stack([y[2500:], X[2500:]]), delimiter=",", fmt="%.5f")
+    model_dir = tmp_path / "model"
+    model_dir.mkdir()
+    return {"train": str(train_dir), "validation": str(val_dir), "model": str(model_dir),
+            "tmp": tmp_path}
+
+
+def _free_port():
+    import socket
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+class TestRunnerCPU:
+    def _run(self, channels, num_gpus, hp_extra=None):
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(_free_port())
+        try:
+            hp = {"num_round": 5, "objective": "binary:logistic", "max_depth": "3",
+                  "eval_metric": "logloss"}
+            hp.update(hp_extra or {})
+            run_training_with_rccl(
+                hyperparameters=hp,
+                train_path=channels["train"],
+                validation_path=channels["validation"],
+                model_dir=channels["model"],
+                content_type="csv",
+                sm_hosts=["algo-1"],
+                current_host="algo-1",
+                checkpoint_dir=None,
+                num_gpus=num_gpus,
+            )
+        finally:
+            os.environ.pop("MASTER_ADDR", None)
+            os.environ.pop("MASTER_PORT", None)
+
+    def test_single_worker_end_to_end(self, channels):
+        self._run(channels, num_gpus=1)
+        model = os.path.join(channels["model"], "xgboost-model")
+        assert os.path.exists(model)
+        from sagemaker_xgboost_container_amd.models.booster import Booster
+
+        b = Booster()
+        b.load_model(model)
+        assert len(b.trees) == 5
+        p = b.predict(np.zeros((2, 6), dtype=np.float32))
+        assert p.shape == (2,)
+
+    def test_two_workers_with_validation_no_deadlock(self, channels):
+        # 2 spawned ranks over gloo; validation is row-sharded so every
+        # rank has an eval set and the fused metric allreduce matches
+        self._run(channels, num_gpus=2)
+        model = os.path.join(channels["model"], "xgboost-model")
+        assert os.path.exists(model)
+        from sagemaker_xgboost_container_amd.models.booster import Booster
+
+        b = Booster()
+        b.load_model(model)
+        assert len(b.trees) == 5
+
+    def test_two_workers_early_stopping(self, channels):
+        # early stopping must fire identically on every rank (aggregated
+        # metrics are rank-identical) — a mismatch would hang the spawn
+        self._run(channels, num_gpus=2,
+                  hp_extra={"early_stopping_rounds": 2, "num_round": 30})
+        assert os.path.exists(os.path.join(channels["model"], "xgboost-model"))
+
+
+class TestRunnerValidation:
+    def test_validation_rules_match_reference(self):
+        # reference distributed_gpu_training.py:60-85 rules
+        errs = validate_gpu_train_configuration("exact", 1, 1, "File", "csv", {})
+        assert any("tree_method" in e for e in errs)
+        errs = validate_gpu_train_configuration("hist", 1, 0, "File", "csv", {})
+        assert any("no GPUs" in e for e in errs)
+        errs = validate_gpu_train_configuration("gpu_hist", 1, 1, "Pipe", "csv", {})
+        assert any("File input mode" in e for e in errs)
+        errs = validate_gpu_train_configuration("hist", 1, 1, "File", "libsvm", {})
+        assert any("input formats" in e for e in errs)
+        errs = validate_gpu_train_configuration(
+            "hist", 2, 1, "File", "csv",
+            {"train": {"S3DistributionType": "ShardedByS3Key"}},
+        )
+        assert any("FullyReplicated" in e for e in errs)
+        assert validate_gpu_train_configuration("hist", 1, 1, "File", "csv", {}) == []
