@@ -118,3 +118,36 @@ class TestRunnerValidation:
         )
         assert any("FullyReplicated" in e for e in errs)
         assert validate_gpu_train_configuration("hist", 1, 1, "File", "csv", {}) == []
+
+
+@pytest.mark.gpu
+class TestRunnerGPU:
+    def test_single_gpu_runner_end_to_end(self, channels):
+        import torch
+
+        assert torch.cuda.is_available()
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(_free_port())
+        try:
+            run_training_with_rccl(
+                hyperparameters={"num_round": 4, "objective": "binary:logistic",
+                                 "max_depth": "4", "eval_metric": "logloss"},
+                train_path=channels["train"],
+                validation_path=channels["validation"],
+                model_dir=channels["model"],
+                content_type="csv",
+                sm_hosts=["algo-1"],
+                current_host="algo-1",
+                checkpoint_dir=None,
+                num_gpus=1,
+            )
+        finally:
+            os.environ.pop("MASTER_ADDR", None)
+            os.environ.pop("MASTER_PORT", None)
+        model = os.path.join(channels["model"], "xgboost-model")
+        assert os.path.exists(model)
+        from sagemaker_xgboost_container_amd.models.booster import Booster
+
+        b = Booster()
+        b.load_model(model)
+        assert len(b.trees) == 4
