@@ -240,6 +240,14 @@ class DMatrix:
                 weight = file_weight
             data = features
 
+        # pandas DataFrame/Series: numeric conversion + column names
+        if hasattr(data, "to_numpy") and hasattr(data, "columns"):
+            if self.feature_names is None:
+                self.feature_names = [str(c) for c in data.columns]
+            data = data.to_numpy(dtype=np.float32)
+        if label is not None and hasattr(label, "to_numpy"):
+            label = label.to_numpy(dtype=np.float32)
+
         if sp.issparse(data):
             self._csr = data.tocsr().astype(np.float32)
         else:

@@ -608,6 +608,20 @@ class Booster:
         return state
 
     def __setstate__(self, state):
+        # pickles of upstream xgboost.core.Booster carry the serialized
+        # model as state["handle"] (raw bytes); with the xgboost shim
+        # installed this class IS xgboost.core.Booster, so route those
+        # through the legacy/JSON byte loader instead of __dict__ update
+        raw = state.get("handle")
+        if raw is not None and isinstance(raw, (bytes, bytearray)) and "params" not in state:
+            from .legacy_binary import load_model_bytes
+
+            self.__init__()
+            load_model_bytes(bytes(raw), booster=self)
+            fnames = state.get("feature_names")
+            if fnames:
+                self.feature_names = list(fnames)
+            return
         self.__dict__.update(state)
 
     def copy(self):

@@ -99,3 +99,56 @@ def test_algorithm_mode_when_no_program(script_env):
     assert env.user_entry_point is None
     training.train(env)
     assert (script_env / "model" / "xgboost-model").exists()
+
+
+XGB_IMPORT_SCRIPT = '''
+"""User script in the reference's public style: `import xgboost as xgb`
+(satisfied by the container's xgboost shim over the native framework)."""
+import argparse
+import os
+
+import xgboost as xgb
+from sagemaker_xgboost_container_amd.data.data_utils import get_dmatrix
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--num_round", type=int, default=5)
+    parser.add_argument("--max_depth", type=int, default=3)
+    args, _ = parser.parse_known_args()
+
+    dtrain = get_dmatrix(os.environ["SM_CHANNEL_TRAIN"], "libsvm")
+    booster = xgb.train(
+        {"objective": "reg:squarederror", "max_depth": args.max_depth},
+        dtrain,
+        num_boost_round=args.num_round,
+    )
+    assert xgb.rabit.get_rank() == 0
+    booster.save_model(os.path.join(os.environ["SM_MODEL_DIR"], "xgboost-model"))
+
+
+if __name__ == "__main__":
+    main()
+'''
+
+
+def test_script_mode_user_script_imports_xgboost(script_env, tmp_path, monkeypatch):
+    """Reference customer scripts depend on `import xgboost as xgb`
+    (abalone_distributed.py, single_machine_customer_script.py); the shim
+    package must satisfy them inside script mode."""
+    code_dir = tmp_path / "code_xgb"
+    code_dir.mkdir()
+    (code_dir / "train_xgb.py").write_text(XGB_IMPORT_SCRIPT)
+    hp_file = script_env / "input/config/hyperparameters.json"
+    hp = json.loads(hp_file.read_text())
+    hp["sagemaker_program"] = "train_xgb.py"
+    hp["sagemaker_submit_directory"] = str(code_dir)
+    hp_file.write_text(json.dumps(hp))
+
+    env = sm_env.TrainingEnv(base_path=str(script_env))
+    training.train(env)
+    model = script_env / "model" / "xgboost-model"
+    assert model.exists()
+    bst = Booster()
+    bst.load_model(str(model))
+    assert len(bst.trees) == 5
