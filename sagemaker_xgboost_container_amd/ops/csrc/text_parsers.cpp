@@ -261,11 +261,13 @@ torch::Tensor parse_csv(const std::string& text, const std::string& delimiter, i
 
   const float nan_v = std::numeric_limits<float>::quiet_NaN();
   std::vector<std::thread> threads;
+  std::atomic<int64_t> bad_fields{0};
   for (int64_t t = 0; t < nt; ++t) {
     threads.emplace_back([&, t] {
       const char* p = data + bounds[t];
       const char* end = data + bounds[t + 1];
       int64_t row = start_row[t];
+      int64_t bad = 0;
       while (p < end) {
         while (p < end && (*p == '\n' || *p == '\r')) ++p;
         if (p >= end) break;
@@ -277,6 +279,7 @@ torch::Tensor parse_csv(const std::string& text, const std::string& delimiter, i
           } else {
             char* next = nullptr;
             rp[c] = std::strtof(p, &next);
+            if (next == p) ++bad;  // non-empty, non-numeric field
             p = (next == p) ? p : next;
             while (p < end && *p != delim && *p != '\n') ++p;  // trailing junk
           }
@@ -285,9 +288,15 @@ torch::Tensor parse_csv(const std::string& text, const std::string& delimiter, i
         while (p < end && *p != '\n') ++p;
         ++row;
       }
+      bad_fields += bad;
     });
   }
   for (auto& th : threads) th.join();
+  if (bad_fields.load() > 0) {
+    throw std::runtime_error(
+        "could not convert " + std::to_string(bad_fields.load()) +
+        " csv field(s) to float (non-numeric data)");
+  }
   return out;
 }
 
