@@ -151,3 +151,53 @@ class TestRunnerGPU:
         b = Booster()
         b.load_model(model)
         assert len(b.trees) == 4
+
+
+class TestAlgorithmModeDispatch:
+    def test_use_dask_gpu_training_dispatches_to_rccl_runner(self, channels, monkeypatch):
+        """`use_dask_gpu_training=true` (reference HP name kept) must route
+        sagemaker_train through the RCCL runner (reference train.py:183-214
+        routed to Dask). Faked SM_NUM_GPUS=2 on CPU exercises the real
+        2-worker spawn end to end."""
+        from sagemaker_xgboost_container_amd.algorithm_mode import train as am_train
+
+        monkeypatch.setenv("SM_NUM_GPUS", "2")
+        monkeypatch.setenv("MASTER_ADDR", "127.0.0.1")
+        monkeypatch.setenv("MASTER_PORT", str(_free_port()))
+        am_train.sagemaker_train(
+            train_config={"num_round": "4", "objective": "binary:logistic",
+                          "max_depth": "3", "tree_method": "hist",
+                          "use_dask_gpu_training": "true"},
+            data_config={"train": {"ContentType": "text/csv", "TrainingInputMode": "File",
+                                   "S3DistributionType": "FullyReplicated"},
+                         "validation": {"ContentType": "text/csv", "TrainingInputMode": "File",
+                                        "S3DistributionType": "FullyReplicated"}},
+            train_path=channels["train"],
+            val_path=channels["validation"],
+            model_dir=channels["model"],
+            sm_hosts=["algo-1"],
+            sm_current_host="algo-1",
+            checkpoint_config={},
+        )
+        model = os.path.join(channels["model"], "xgboost-model")
+        assert os.path.exists(model)
+
+    def test_use_dask_gpu_training_validation_errors(self, channels, monkeypatch):
+        from sagemaker_xgboost_container_amd.algorithm_mode import train as am_train
+        from sagemaker_xgboost_container_amd.toolkit import exceptions as exc
+
+        monkeypatch.setenv("SM_NUM_GPUS", "0")  # no GPUs -> UserError
+        with pytest.raises(exc.UserError) as ei:
+            am_train.sagemaker_train(
+                train_config={"num_round": "4", "objective": "binary:logistic",
+                              "use_dask_gpu_training": "true"},
+                data_config={"train": {"ContentType": "text/csv", "TrainingInputMode": "File",
+                                       "S3DistributionType": "FullyReplicated"}},
+                train_path=channels["train"],
+                val_path=None,
+                model_dir=channels["model"],
+                sm_hosts=["algo-1"],
+                sm_current_host="algo-1",
+                checkpoint_config={},
+            )
+        assert "unsuitable for multi-GPU" in str(ei.value)
