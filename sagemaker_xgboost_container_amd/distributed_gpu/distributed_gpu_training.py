@@ -67,8 +67,15 @@ def _worker(local_rank, num_gpus, sm_hosts, current_host, hyperparameters, train
     master_port = os.environ.get("MASTER_PORT", "23456")
 
     # nccl(=RCCL) on GPU instances; gloo keeps the same code path testable
-    # (and running) on CPU-only hosts
-    use_cuda = torch.cuda.is_available()
+    # (and running) on CPU-only hosts. RCCL refuses two ranks on one
+    # device, so an oversubscribed request (more workers than GPUs) also
+    # falls back to gloo/CPU instead of crashing.
+    use_cuda = torch.cuda.is_available() and num_gpus <= torch.cuda.device_count()
+    if torch.cuda.is_available() and not use_cuda:
+        logger.warning(
+            "%d workers requested but only %d GPU(s) present; running over gloo on CPU",
+            num_gpus, torch.cuda.device_count(),
+        )
     if use_cuda:
         torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
     dist.init_process_group(
