@@ -583,7 +583,7 @@ __global__ __launch_bounds__(BLOCK) void hist_device_kernel(
     const LevelNode* __restrict__ nodes, const int* __restrict__ hist_prefix,
     const LevelWork* __restrict__ work, unsigned long long* __restrict__ out,
     int k, int nfeat, int stride, int n_groups, int feats_per_group,
-    const float* __restrict__ gh_max, int rows_per_block) {
+    const float* __restrict__ gh_max, int rows_per_block, int slot_lo, int slot_hi) {
   extern __shared__ unsigned long long lhist[];
   const float scale_g = 8589934592.0f / fmaxf(gh_max[0], 1e-30f);
   const float scale_h = 8589934592.0f / fmaxf(gh_max[1], 1e-30f);
@@ -593,6 +593,7 @@ __global__ __launch_bounds__(BLOCK) void hist_device_kernel(
     const int fg = vb % n_groups;
     const int hvb = vb / n_groups;
     const int slot = find_slot(hist_prefix, k, hvb);
+    if (slot < slot_lo || slot >= slot_hi) continue;  // comm-overlap slot window
     const int chunk = hvb - hist_prefix[slot];
     const int nb = hist_prefix[slot + 1] - hist_prefix[slot];
     const LevelNode node = nodes[slot];
@@ -1409,7 +1410,7 @@ static void launch_hist_device(int grid, int hist_block, size_t lds_bytes, hipSt
                                const int* hist_prefix, const LevelWork* work,
                                unsigned long long* acc, int k, int nfeat, int stride,
                                int n_groups, int feats_per_group, const float* gh_max,
-                               int rows_per_block) {
+                               int rows_per_block, int slot_lo = 0, int slot_hi = 1 << 30) {
   // hipGraph kernel nodes validate dynamic LDS against the function's
   // max-dynamic-shared attribute (plain launches do not): without this
   // opt-in a captured full-slab launch (>64 KB) reads a truncated LDS
@@ -1425,11 +1426,12 @@ static void launch_hist_device(int grid, int hist_block, size_t lds_bytes, hipSt
   if (hist_block == 512) {
     hipLaunchKernelGGL((hist_device_kernel<BinT, 512>), dim3(grid), dim3(512), lds_bytes, stream,
                        bins_c, gh_c, nodes, hist_prefix, work, acc, k, nfeat, stride, n_groups,
-                       feats_per_group, gh_max, rows_per_block);
+                       feats_per_group, gh_max, rows_per_block, slot_lo, slot_hi);
   } else {
     hipLaunchKernelGGL((hist_device_kernel<BinT, HIST_BLOCK>), dim3(grid), dim3(HIST_BLOCK),
                        lds_bytes, stream, bins_c, gh_c, nodes, hist_prefix, work, acc, k, nfeat,
-                       stride, n_groups, feats_per_group, gh_max, rows_per_block);
+                       stride, n_groups, feats_per_group, gh_max, rows_per_block, slot_lo,
+                       slot_hi);
   }
 }
 
@@ -1437,7 +1439,8 @@ void grow_hist_level(torch::Tensor bins_c, torch::Tensor gh_c, torch::Tensor nod
                      torch::Tensor hist_prefix, torch::Tensor work, torch::Tensor acc,
                      int64_t k, int64_t nfeat, int64_t stride, int64_t n_groups,
                      int64_t feats_per_group, torch::Tensor gh_max, int64_t rows_per_block,
-                     int64_t grid, int64_t lds_words, int64_t hist_block) {
+                     int64_t grid, int64_t lds_words, int64_t hist_block,
+                     int64_t slot_lo, int64_t slot_hi) {
   const size_t lds_bytes = (size_t)lds_words * sizeof(unsigned long long);
   auto stream = current_stream();
   if (bins_c.scalar_type() == torch::kUInt8) {
@@ -1446,14 +1449,16 @@ void grow_hist_level(torch::Tensor bins_c, torch::Tensor gh_c, torch::Tensor nod
         (const float2*)gh_c.data_ptr<float>(), (const LevelNode*)nodes.data_ptr<int>(),
         hist_prefix.data_ptr<int>(), (const LevelWork*)work.data_ptr<int>(),
         (unsigned long long*)acc.data_ptr<int64_t>(), (int)k, (int)nfeat, (int)stride,
-        (int)n_groups, (int)feats_per_group, gh_max.data_ptr<float>(), (int)rows_per_block);
+        (int)n_groups, (int)feats_per_group, gh_max.data_ptr<float>(), (int)rows_per_block,
+        (int)slot_lo, (int)slot_hi);
   } else {
     launch_hist_device<short>(
         (int)grid, (int)hist_block, lds_bytes, stream, bins_c.data_ptr<short>(),
         (const float2*)gh_c.data_ptr<float>(), (const LevelNode*)nodes.data_ptr<int>(),
         hist_prefix.data_ptr<int>(), (const LevelWork*)work.data_ptr<int>(),
         (unsigned long long*)acc.data_ptr<int64_t>(), (int)k, (int)nfeat, (int)stride,
-        (int)n_groups, (int)feats_per_group, gh_max.data_ptr<float>(), (int)rows_per_block);
+        (int)n_groups, (int)feats_per_group, gh_max.data_ptr<float>(), (int)rows_per_block,
+        (int)slot_lo, (int)slot_hi);
   }
 }
 

@@ -798,26 +798,64 @@ class DeviceGrower:
             )
             acc_d = self.acc[:k]
             acc_d.zero_()
-            _K.grow_hist_level(
-                src_bins, src_gh, nodes_d, self.hp[d], self.work[d], acc_d,
-                k, f, stride, self.n_groups, self.feats_per_group, scale,
-                self.rows_per_block, _GROW_HIST_GRID, self.lds_words, self.hist_block,
+            overlap = (
+                comm is not None
+                and k >= 4
+                and _os.environ.get("SMXGB_COMM_OVERLAP") == "1"
+                and hasattr(comm, "allreduce_async_")
             )
-            if comm is not None:
-                if k == 1:
-                    comm.allreduce_(acc_d)
-                else:
-                    # Compacted allreduce: of each sibling pair exactly one
-                    # slot was built from rows (the other is all zeros —
-                    # make_level's subtraction trick), so folding pairs
-                    # halves the message without a gather kernel. The sum
-                    # lands back in BOTH child slots; convert_level only
-                    # reads the built one. int64 adds wrap mod 2^64,
-                    # matching the kernel's unsigned fixed-point.
-                    compact = acc_d[0::2] + acc_d[1::2]
-                    comm.allreduce_(compact)
-                    acc_d[0::2] = compact
-                    acc_d[1::2] = compact
+            if overlap:
+                # Chunked hist/allreduce pipelining: the collective for the
+                # first half of the level's slots runs on the comm stream
+                # WHILE the compute stream builds the second half's
+                # histograms (the nccl stream only waits on the tensors it
+                # reduces). Fold-compaction per half as below.
+                mid = k >> 1  # even: pairs stay within a half
+                _K.grow_hist_level(
+                    src_bins, src_gh, nodes_d, self.hp[d], self.work[d], acc_d,
+                    k, f, stride, self.n_groups, self.feats_per_group, scale,
+                    self.rows_per_block, _GROW_HIST_GRID, self.lds_words, self.hist_block,
+                    0, mid,
+                )
+                c1 = acc_d[0:mid:2] + acc_d[1:mid:2]
+                w1 = comm.allreduce_async_(c1)
+                _K.grow_hist_level(
+                    src_bins, src_gh, nodes_d, self.hp[d], self.work[d], acc_d,
+                    k, f, stride, self.n_groups, self.feats_per_group, scale,
+                    self.rows_per_block, _GROW_HIST_GRID, self.lds_words, self.hist_block,
+                    mid, k,
+                )
+                c2 = acc_d[mid::2] + acc_d[mid + 1 :: 2]
+                w2 = comm.allreduce_async_(c2)
+                w1.wait()
+                acc_d[0:mid:2] = c1
+                acc_d[1:mid:2] = c1
+                w2.wait()
+                acc_d[mid::2] = c2
+                acc_d[mid + 1 :: 2] = c2
+            else:
+                _K.grow_hist_level(
+                    src_bins, src_gh, nodes_d, self.hp[d], self.work[d], acc_d,
+                    k, f, stride, self.n_groups, self.feats_per_group, scale,
+                    self.rows_per_block, _GROW_HIST_GRID, self.lds_words, self.hist_block,
+                    0, 1 << 30,
+                )
+                if comm is not None:
+                    if k == 1:
+                        comm.allreduce_(acc_d)
+                    else:
+                        # Compacted allreduce: of each sibling pair exactly
+                        # one slot was built from rows (the other is all
+                        # zeros — make_level's subtraction trick), so
+                        # folding pairs halves the message without a gather
+                        # kernel. The sum lands back in BOTH child slots;
+                        # convert_level only reads the built one. int64
+                        # adds wrap mod 2^64, matching the kernel's
+                        # unsigned fixed-point.
+                        compact = acc_d[0::2] + acc_d[1::2]
+                        comm.allreduce_(compact)
+                        acc_d[0::2] = compact
+                        acc_d[1::2] = compact
             hist_d = self.hist_f32[base : base + k]
             _K.grow_convert_level(acc_d, hist_d, nodes_d, k, self.slots2, scale)
             if d > 0:

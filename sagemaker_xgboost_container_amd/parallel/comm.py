@@ -28,6 +28,12 @@ class Communicator:
         dist.all_reduce(tensor, op=dist.ReduceOp.SUM, group=self.group)
         return tensor
 
+    def allreduce_async_(self, tensor):
+        """Non-blocking sum-allreduce; returns a work handle (`.wait()`).
+        On nccl the collective runs on the comm stream, so compute kernels
+        enqueued after this call overlap with it until wait()."""
+        return dist.all_reduce(tensor, op=dist.ReduceOp.SUM, group=self.group, async_op=True)
+
     def allreduce_max_(self, tensor):
         dist.all_reduce(tensor, op=dist.ReduceOp.MAX, group=self.group)
         return tensor
