@@ -92,11 +92,23 @@ def _single_rank_worker(port, q):
         bst_ref = trainer.train(
             params, DMatrix(X, label=y), num_boost_round=5, verbose_eval=False,
         )
-        same_trees = json.dumps(
-            bst.save_json()["learner"]["gradient_booster"]["model"]["trees"], sort_keys=True
-        ) == json.dumps(
-            bst_ref.save_json()["learner"]["gradient_booster"]["model"]["trees"], sort_keys=True
-        )
+
+        def _trees(b):
+            return json.dumps(
+                b.save_json()["learner"]["gradient_booster"]["model"]["trees"], sort_keys=True
+            )
+
+        same_trees = _trees(bst) == _trees(bst_ref)
+
+        # chunked-overlap variant through REAL nccl async allreduce handles
+        os.environ["SMXGB_COMM_OVERLAP"] = "1"
+        try:
+            bst_ov = trainer.train(
+                params, DMatrix(X, label=y), num_boost_round=5, verbose_eval=False, comm=comm,
+            )
+        finally:
+            os.environ.pop("SMXGB_COMM_OVERLAP", None)
+        same_trees = same_trees and (_trees(bst_ov) == _trees(bst_ref))
         q.put(("ok", res["train"]["logloss"][-1], same_trees))
         dist.barrier()
         dist.destroy_process_group()
