@@ -798,11 +798,17 @@ class DeviceGrower:
             )
             acc_d = self.acc[:k]
             acc_d.zero_()
+            # Default ON for real multi-rank groups (the driver's scaling
+            # run): correctness is proven bit-identical under gloo 2-rank
+            # and nccl async handles (tests/test_distributed_gpu.py,
+            # tests/test_rccl_rehearsal.py). SMXGB_COMM_OVERLAP=0 disables,
+            # =1 forces even at world 1 (rehearsal).
+            _ov_env = _os.environ.get("SMXGB_COMM_OVERLAP")
             overlap = (
                 comm is not None
                 and k >= 4
-                and _os.environ.get("SMXGB_COMM_OVERLAP") == "1"
                 and hasattr(comm, "allreduce_async_")
+                and (_ov_env == "1" or (_ov_env != "0" and getattr(comm, "world_size", 1) > 1))
             )
             if overlap:
                 # Chunked hist/allreduce pipelining: the collective for the
