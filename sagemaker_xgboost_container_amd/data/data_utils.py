@@ -344,6 +344,8 @@ def get_recordio_protobuf_dmatrix(path, is_pipe=False):
 
 def get_dmatrix(data_path, content_type, csv_weights=0, is_pipe=False):
     """Build a DMatrix for a channel path; None if the path has no data."""
+    if data_path is None:
+        return None
     if is_pipe:
         files_path = _get_pipe_mode_files_path(data_path)
     else:

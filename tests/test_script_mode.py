@@ -152,3 +152,49 @@ def test_script_mode_user_script_imports_xgboost(script_env, tmp_path, monkeypat
     bst = Booster()
     bst.load_model(str(model))
     assert len(bst.trees) == 5
+
+
+REF_ABALONE_SCRIPT = "/root/reference/test/resources/abalone/abalone_distributed.py"
+
+
+@pytest.mark.skipif(not os.path.exists(REF_ABALONE_SCRIPT), reason="reference fixtures absent")
+def test_reference_abalone_script_runs_unmodified(script_env, tmp_path):
+    """The reference's PUBLIC script-mode example (abalone_distributed.py)
+    must run VERBATIM: it imports `xgboost`, `sagemaker_containers.
+    entry_point` and `sagemaker_xgboost_container.{distributed,data_utils}`
+    — all satisfied by this container's compatibility packages."""
+    code_dir = tmp_path / "ref_code"
+    code_dir.mkdir()
+    (code_dir / "abalone_distributed.py").write_text(open(REF_ABALONE_SCRIPT).read())
+
+    # the reference example also reads a validation channel
+    rng = np.random.default_rng(1)
+    Xv = rng.normal(size=(100, 6))
+    yv = Xv[:, 0] * 2
+    vlines = [
+        f"{yv[i]:.5f} " + " ".join(f"{j}:{Xv[i, j]:.5f}" for j in range(6)) for i in range(100)
+    ]
+    (script_env / "input/data/validation/data.libsvm").write_text("\n".join(vlines))
+    os.environ[smc.SM_CHANNEL_VALIDATION] = str(script_env / "input/data/validation")
+
+    hp_file = script_env / "input/config/hyperparameters.json"
+    hp = {
+        "sagemaker_program": "abalone_distributed.py",
+        "sagemaker_submit_directory": str(code_dir),
+        "num_round": "5",
+        "max_depth": "3",
+        "eta": "0.3",
+        "gamma": "0",
+        "min_child_weight": "1",
+        "subsample": "0.9",
+        "objective": "reg:squarederror",
+    }
+    hp_file.write_text(json.dumps(hp))
+
+    env = sm_env.TrainingEnv(base_path=str(script_env))
+    training.train(env)
+    model = script_env / "model" / "xgboost-model"
+    assert model.exists()
+    bst = Booster()
+    bst.load_model(str(model))
+    assert len(bst.trees) == 5
