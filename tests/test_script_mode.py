@@ -158,7 +158,7 @@ REF_ABALONE_SCRIPT = "/root/reference/test/resources/abalone/abalone_distributed
 
 
 @pytest.mark.skipif(not os.path.exists(REF_ABALONE_SCRIPT), reason="reference fixtures absent")
-def test_reference_abalone_script_runs_unmodified(script_env, tmp_path):
+def test_reference_abalone_script_runs_unmodified(script_env, tmp_path, monkeypatch):
     """The reference's PUBLIC script-mode example (abalone_distributed.py)
     must run VERBATIM: it imports `xgboost`, `sagemaker_containers.
     entry_point` and `sagemaker_xgboost_container.{distributed,data_utils}`
@@ -175,7 +175,7 @@ def test_reference_abalone_script_runs_unmodified(script_env, tmp_path):
         f"{yv[i]:.5f} " + " ".join(f"{j}:{Xv[i, j]:.5f}" for j in range(6)) for i in range(100)
     ]
     (script_env / "input/data/validation/data.libsvm").write_text("\n".join(vlines))
-    os.environ[smc.SM_CHANNEL_VALIDATION] = str(script_env / "input/data/validation")
+    monkeypatch.setenv(smc.SM_CHANNEL_VALIDATION, str(script_env / "input/data/validation"))
 
     hp_file = script_env / "input/config/hyperparameters.json"
     hp = {
