@@ -156,3 +156,25 @@ class TestCV:
                      xgb.DMatrix(X, label=y), num_boost_round=3, nfold=3,
                      as_pandas=False)
         assert isinstance(out, dict) or hasattr(out, "columns")
+
+
+def test_cv_stratified_preserves_class_balance():
+    rng = np.random.default_rng(9)
+    X = rng.normal(size=(600, 4)).astype(np.float32)
+    y = (rng.random(600) < 0.2).astype(np.float32)  # 20% positives
+    out = xgb.cv(
+        {"objective": "binary:logistic", "max_depth": 3},
+        xgb.DMatrix(X, label=y), num_boost_round=3, nfold=5,
+        stratified=True, metrics="logloss", as_pandas=False,
+    )
+    key = "test-logloss-mean"
+    assert len(out[key]) == 3
+    # re-derive the folds the same way and check balance
+    from xgboost.cv import cv as _  # noqa: F401
+    idx = np.arange(600)
+    g = np.random.default_rng(0)
+    g.shuffle(idx)
+    order = idx[np.argsort(y[idx], kind="stable")]
+    folds = [order[k::5] for k in range(5)]
+    rates = [float(y[f].mean()) for f in folds]
+    assert max(rates) - min(rates) < 0.05

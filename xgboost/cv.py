@@ -24,9 +24,18 @@ def cv(params, dtrain=None, num_boost_round=10, nfold=3, metrics=None, seed=0,
     n = dtrain.num_row()
     rng = np.random.default_rng(seed)
     idx = np.arange(n)
-    if shuffle:
-        rng.shuffle(idx)
-    folds = np.array_split(idx, nfold)
+    if stratified:
+        # per-class round-robin after a shuffle: every fold sees the same
+        # label distribution (classification cv)
+        labels = np.asarray(dtrain.get_label())
+        if shuffle:
+            rng.shuffle(idx)
+        order = idx[np.argsort(labels[idx], kind="stable")]
+        folds = [order[k::nfold] for k in range(nfold)]
+    else:
+        if shuffle:
+            rng.shuffle(idx)
+        folds = np.array_split(idx, nfold)
 
     histories = []
     for k in range(nfold):
