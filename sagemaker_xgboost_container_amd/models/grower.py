@@ -286,10 +286,16 @@ class HistGrower:
         if streams is None:
             streams = {}
             self._streams = streams
-        s = streams.get(slot)
+        import os as _os
+
+        # SMXGB_CLASS_STREAMS caps concurrent streams (slots share
+        # round-robin); default one stream per slot
+        cap = int(_os.environ.get("SMXGB_CLASS_STREAMS", "0") or 0)
+        key = slot % cap if cap > 0 else slot
+        s = streams.get(key)
         if s is None:
             s = torch.cuda.Stream(device=self.device)
-            streams[slot] = s
+            streams[key] = s
         return s
 
     def grow_async(self, gh, slot=0):
