@@ -74,6 +74,16 @@ class XGBModel:
                 (DMatrix(_as_array(ex), label=np.asarray(ey, dtype=np.float32)), f"validation_{i}")
                 for i, (ex, ey) in enumerate(eval_set)
             ]
+        maximize = None
+        if early_stopping_rounds:
+            from sagemaker_xgboost_container_amd.constants.xgb_constants import (
+                XGB_MAXIMIZE_METRICS,
+            )
+
+            metric = params.get("eval_metric")
+            if isinstance(metric, (list, tuple)):
+                metric = metric[-1] if metric else None
+            maximize = bool(metric) and str(metric).split("@")[0] in XGB_MAXIMIZE_METRICS
         self.evals_result_ = {}
         self._booster = _train(
             params,
@@ -83,6 +93,7 @@ class XGBModel:
             evals_result=self.evals_result_,
             verbose_eval=verbose,
             early_stopping_rounds=early_stopping_rounds,
+            maximize=maximize,
             xgb_model=xgb_model,
         )
         return self
