@@ -1208,17 +1208,30 @@ __global__ __launch_bounds__(HIST_BLOCK) void leaf_update_kernel(
 // batched forest prediction (dense rows x trees traversal)
 // ---------------------------------------------------------------------------
 
+// Work item = (row, tree-chunk): small serving batches (1k rows) would
+// otherwise launch ~4 blocks against 256 CUs and serialize 500 trees per
+// thread (measured 1.36 ms for 1k x 500). Each (row, chunk) owns its
+// exclusive out[chunk, row, :] slice — no atomics; the caller reduces the
+// chunk axis with a fixed-order torch sum (deterministic).
 __global__ __launch_bounds__(HIST_BLOCK) void predict_kernel(
     const float* __restrict__ X, long long n, int nfeat,
     const int* __restrict__ left, const int* __restrict__ right,
     const int* __restrict__ feat, const float* __restrict__ thresh,
     const unsigned char* __restrict__ defl, const float* __restrict__ value,
     const int* __restrict__ tree_root, const int* __restrict__ tree_cls,
-    int t_begin, int t_end, float* __restrict__ out, int k) {
+    int t_begin, int t_end, float* __restrict__ out, int k, int n_chunks,
+    int trees_per_chunk) {
+  const long long total = n * (long long)n_chunks;
   const long long step = (long long)gridDim.x * blockDim.x;
-  for (long long row = (long long)blockIdx.x * blockDim.x + threadIdx.x; row < n; row += step) {
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += step) {
+    const long long row = idx / n_chunks;
+    const int c = (int)(idx - row * n_chunks);
     const float* xr = X + row * nfeat;
-    for (int t = t_begin; t < t_end; ++t) {
+    float* orow = out + ((long long)c * n + row) * k;
+    const int ts = t_begin + c * trees_per_chunk;
+    const int te = min(t_end, ts + trees_per_chunk);
+    for (int t = ts; t < te; ++t) {
       int nid = tree_root[t];
       int l;
       while ((l = left[nid]) >= 0) {
@@ -1226,7 +1239,7 @@ __global__ __launch_bounds__(HIST_BLOCK) void predict_kernel(
         const bool goleft = isnan(fv) ? (defl[nid] != 0) : (fv < thresh[nid]);
         nid = goleft ? l : right[nid];
       }
-      out[row * k + tree_cls[t]] += value[nid];
+      orow[tree_cls[t]] += value[nid];
     }
   }
 }
@@ -1303,16 +1316,19 @@ void leaf_update(torch::Tensor buf0, torch::Tensor buf1, torch::Tensor margin_ba
 void predict_forest(torch::Tensor X, torch::Tensor left, torch::Tensor right,
                     torch::Tensor feat, torch::Tensor thresh, torch::Tensor defl,
                     torch::Tensor value, torch::Tensor tree_root, torch::Tensor tree_cls,
-                    int64_t t_begin, int64_t t_end, torch::Tensor out, int64_t k) {
+                    int64_t t_begin, int64_t t_end, torch::Tensor out, int64_t k,
+                    int64_t n_chunks, int64_t trees_per_chunk) {
   CHECK_GPU(X);
   const long long n = X.size(0);
-  const int grid = (int)std::min<long long>((n + HIST_BLOCK - 1) / HIST_BLOCK, 2048);
+  const long long total = n * n_chunks;
+  const int grid = (int)std::min<long long>((total + HIST_BLOCK - 1) / HIST_BLOCK, 4096);
   hipLaunchKernelGGL(predict_kernel, dim3(std::max(grid, 1)), dim3(HIST_BLOCK), 0, current_stream(),
                      X.data_ptr<float>(), n, (int)X.size(1), left.data_ptr<int>(),
                      right.data_ptr<int>(), feat.data_ptr<int>(), thresh.data_ptr<float>(),
                      defl.data_ptr<unsigned char>(), value.data_ptr<float>(),
                      tree_root.data_ptr<int>(), tree_cls.data_ptr<int>(),
-                     (int)t_begin, (int)t_end, out.data_ptr<float>(), (int)k);
+                     (int)t_begin, (int)t_end, out.data_ptr<float>(), (int)k, (int)n_chunks,
+                     (int)trees_per_chunk);
 }
 
 void hist_build_compact(torch::Tensor bins_c, torch::Tensor gh_c, torch::Tensor jobs,

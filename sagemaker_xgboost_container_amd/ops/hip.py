@@ -298,19 +298,53 @@ class _FlatForest:
         self.n_trees = len(trees)
 
 
+def _predict_chunking(n, T):
+    """(n_chunks, trees_per_chunk): split the tree axis so small batches
+    still produce >=128k work items (a 1k-row, 500-tree request fills 4
+    blocks otherwise — measured 1.36 ms vs the chip's ~0.1 ms)."""
+    target = 131072
+    n_chunks = int(min(T, max(1, (target + n - 1) // max(n, 1))))
+    trees_per_chunk = (T + n_chunks - 1) // n_chunks
+    n_chunks = (T + trees_per_chunk - 1) // trees_per_chunk
+    return n_chunks, trees_per_chunk
+
+
+def _run_predict(X, fl, k, t_begin, t_end):
+    n = X.shape[0]
+    T = t_end - t_begin
+    n_chunks, tpc = _predict_chunking(n, T)
+    if n_chunks > 1:
+        part = torch.zeros((n_chunks, n, k), dtype=torch.float32, device=X.device)
+        _K.predict_forest(
+            X.contiguous(), fl["left"], fl["right"], fl["feature"], fl["threshold"],
+            fl["default_left"], fl["value"], fl["tree_root"], fl["tree_cls"],
+            t_begin, t_end, part.reshape(-1, k), k, n_chunks, tpc,
+        )
+        return part.sum(0)  # fixed-order reduce: deterministic
+    out = torch.zeros((n, k), dtype=torch.float32, device=X.device)
+    _K.predict_forest(
+        X.contiguous(), fl["left"], fl["right"], fl["feature"], fl["threshold"],
+        fl["default_left"], fl["value"], fl["tree_root"], fl["tree_cls"],
+        t_begin, t_end, out, k, 1, T,
+    )
+    return out
+
+
 def predict_forest(trees, tree_info, X, k, t_begin=0, t_end=None, out=None):
     """Summed margin contributions of trees[t_begin:t_end] -> (n, k)."""
     forest = _FlatForest(trees, tree_info, X.device)
-    n = X.shape[0]
-    if out is None:
-        out = torch.zeros((n, k), dtype=torch.float32, device=X.device)
     if t_end is None:
         t_end = forest.n_trees
-    _K.predict_forest(
-        X.contiguous(), forest.left, forest.right, forest.feat, forest.thresh, forest.defl,
-        forest.value, forest.tree_root, forest.tree_cls, t_begin, t_end, out, k,
-    )
-    return out
+    fl = {
+        "left": forest.left, "right": forest.right, "feature": forest.feat,
+        "threshold": forest.thresh, "default_left": forest.defl, "value": forest.value,
+        "tree_root": forest.tree_root, "tree_cls": forest.tree_cls,
+    }
+    res = _run_predict(X, fl, k, t_begin, t_end)
+    if out is not None:
+        out += res
+        return out
+    return res
 
 
 def predict_tree(tree, X):
@@ -344,16 +378,9 @@ def make_flat_forest(trees, tree_info, weight_drop, device):
 def predict_forest_flat(flat, X, k, t_begin=0, t_end=None):
     if t_end is None:
         t_end = flat["n_trees"]
-    n = X.shape[0]
-    out = torch.zeros((n, k), dtype=torch.float32, device=X.device)
     if t_end <= t_begin:
-        return out
-    _K.predict_forest(
-        X.contiguous(), flat["left"], flat["right"], flat["feature"], flat["threshold"],
-        flat["default_left"], flat["value"], flat["tree_root"], flat["tree_cls"],
-        t_begin, t_end, out, k,
-    )
-    return out
+        return torch.zeros((X.shape[0], k), dtype=torch.float32, device=X.device)
+    return _run_predict(X, flat, k, t_begin, t_end)
 
 
 class TreeState:
