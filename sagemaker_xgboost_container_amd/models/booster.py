@@ -211,16 +211,12 @@ class Booster:
                     f"feature_names mismatch: model expects {self.num_features} features, "
                     f"got {arr.shape[1]}"
                 )
+        # with the tree-chunked predict kernel the GPU wins at every batch
+        # size (100x500 trees: 0.085 ms GPU vs 0.70 ms CPU —
+        # benchmarks/bench_predict_cpu_gpu.py), so GPU is always preferred;
+        # predictor=cpu_predictor still forces host traversal
         predictor = self.params.get("predictor")
         use_gpu = torch.cuda.is_available() and predictor != "cpu_predictor"
-        if use_gpu and predictor != "gpu_predictor":
-            # auto: tiny batches are launch-bound on GPU — the parallel C++
-            # traversal wins below ~150k row-trees (measured crossover,
-            # benchmarks/bench_predict_cpu_gpu.py: 100x500 trees CPU 0.70ms
-            # vs GPU 0.92ms; 1000x500 GPU 1.03ms vs CPU 7.2ms)
-            n_rows = arr.shape[0] if hasattr(arr, "shape") else len(arr)
-            if n_rows * max(1, len(self.trees)) < 150_000:
-                use_gpu = False
         return torch.as_tensor(arr, dtype=torch.float32, device="cuda" if use_gpu else "cpu")
 
     def _cpu_flat_forest(self):
