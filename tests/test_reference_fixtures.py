@@ -254,3 +254,60 @@ class TestWeightSidecarFixtures:
         w = dm.get_weight()
         assert len(w) == dm.num_row()
         assert abs(float(w[0]) - 0.2) < 1e-6
+
+
+class TestMixedFormatEnsemble:
+    def test_ensemble_of_legacy_and_json_models(self, tmp_path, monkeypatch):
+        """The reference serves every model file in the dir as an ensemble
+        (serve_utils.py:171-197) — formats may be mixed: an old-binary
+        Booster next to a JSON one must load and average."""
+        import shutil
+
+        from sagemaker_xgboost_container_amd.algorithm_mode import serve_utils
+        from sagemaker_xgboost_container_amd.constants import sm_env_constants as smc
+        from sagemaker_xgboost_container_amd.models.booster import Booster
+
+        src = f"{REF}/models/saved_booster/xgboost-model"
+        shutil.copy(src, tmp_path / "model-a")
+        b = Booster()
+        b.load_model(src)
+        b.save_model(tmp_path / "model-b")  # same model, JSON form
+
+        monkeypatch.setenv(smc.SAGEMAKER_INFERENCE_ENSEMBLE, "true")
+        boosters, formats = serve_utils.get_loaded_booster(str(tmp_path), ensemble=True)
+        assert len(boosters) == 2
+        assert set(formats) == {"xgb_format"}
+
+        from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+
+        X = np.random.default_rng(3).normal(size=(8, 4)).astype(np.float32)
+        preds = serve_utils.predict(boosters, formats, DMatrix(X), "text/csv")
+        single = b.predict(X)
+        # both members are the same model -> ensemble mean == single model
+        np.testing.assert_allclose(np.asarray(preds), single, atol=1e-5)
+
+    def test_mme_serves_legacy_binary_model(self, tmp_path):
+        """Multi-model endpoint: loading a legacy-binary model through the
+        MME management API and invoking it."""
+        import shutil
+
+        from fastapi.testclient import TestClient
+
+        from sagemaker_xgboost_container_amd import serving_mms
+
+        model_dir = tmp_path / "legacy"
+        model_dir.mkdir()
+        shutil.copy(f"{REF}/models/saved_booster/xgboost-model", model_dir / "xgboost-model")
+
+        client = TestClient(serving_mms.app)
+        r = client.post("/models", json={"model_name": "legacy", "url": str(model_dir)})
+        assert r.status_code == 200, r.text
+        r = client.post(
+            "/models/legacy/invoke",
+            content="1,0.5,0.4,0.1",
+            headers={"Content-Type": "text/csv"},
+        )
+        assert r.status_code == 200, r.text
+        vals = [float(v) for v in r.text.strip().split("\n")[0].split(",")]
+        assert len(vals) == 3 and abs(sum(vals) - 1.0) < 1e-4
+        assert client.delete("/models/legacy").status_code == 200
