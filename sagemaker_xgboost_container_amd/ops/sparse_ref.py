@@ -196,7 +196,17 @@ def quantize_sparse(csr, max_bin=256, sample_weight=None, comm=None):
 
 
 def _gather_ranges(starts, counts):
-    """Concatenate [s, s+c) ranges — vectorized multi-range gather."""
+    """Concatenate [s, s+c) ranges — vectorized multi-range gather.
+
+    Zero-count ranges are dropped first: with them present the cumsum
+    boundary trick writes two boundary adjustments into the SAME slot
+    (duplicate cc values) and silently corrupts the gather — empty libsvm
+    rows (label only, no features) hit exactly that.
+    """
+    nz = counts > 0
+    if not bool(nz.all()):
+        starts = starts[nz]
+        counts = counts[nz]
     total = int(counts.sum())
     if total == 0:
         return torch.zeros(0, dtype=torch.int64)

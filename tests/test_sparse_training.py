@@ -250,3 +250,47 @@ class TestDistributedSparse:
         assert len(sigs) == 1, "ranks grew different trees on the sparse path"
         assert results[0][3] == pytest.approx(results[1][3], abs=1e-12)
         assert results[0][3] < 0.6
+
+
+class TestEmptyRows:
+    def test_gather_ranges_with_zero_count_rows(self):
+        import torch
+
+        from sagemaker_xgboost_container_amd.ops.sparse_ref import _gather_ranges
+
+        starts = torch.tensor([10, 20, 30, 40, 50, 60])
+        counts = torch.tensor([2, 0, 3, 0, 0, 1])
+        assert _gather_ranges(starts, counts).tolist() == [10, 11, 30, 31, 32, 60]
+
+    def test_sparse_training_with_empty_rows_matches_dense(self):
+        """libsvm rows with a label but NO features (the reference corpus
+        contains such lines) must train identically to the dense path —
+        the gather previously corrupted histograms around them."""
+        rng = np.random.default_rng(31)
+        n, f = 1500, 80
+        csr, y = _rand_csr(n, f, 0.15, seed=31)
+        # blank out ~10% of rows entirely (label-only rows)
+        empty = rng.choice(n, size=n // 10, replace=False)
+        mask = np.ones(n, dtype=bool)
+        mask[empty] = False
+        lil = csr.tolil()
+        lil[empty, :] = 0
+        csr = lil.tocsr()
+        csr.eliminate_zeros()
+        import copy
+        import json as _json
+
+        outs = {}
+        for mode in ("1", "0"):
+            os.environ["SMXGB_SPARSE"] = mode
+            try:
+                bst = trainer.train(
+                    {"objective": "binary:logistic", "max_depth": 4, "eta": 0.3, "device": "cpu"},
+                    DMatrix(csr.copy(), label=y), num_boost_round=4, verbose_eval=False,
+                )
+            finally:
+                os.environ.pop("SMXGB_SPARSE", None)
+            outs[mode] = _json.dumps(
+                bst.save_json()["learner"]["gradient_booster"]["model"]["trees"], sort_keys=True
+            )
+        assert outs["1"] == outs["0"]
