@@ -48,8 +48,21 @@ class GrowParams:
         self.colsample_bytree = float(p.get("colsample_bytree", 1.0))
         self.colsample_bylevel = float(p.get("colsample_bylevel", 1.0))
         self.colsample_bynode = float(p.get("colsample_bynode", 1.0))
-        self.monotone_constraints = p.get("monotone_constraints")
-        self.interaction_constraints = p.get("interaction_constraints")
+        # xgboost accepts the SageMaker string forms "(1,-1,0)" and
+        # "[[0,1],[2,3]]" directly (the reference forwards hyperparameter
+        # strings as-is) — normalize them here so trainer.train keeps that
+        # surface
+        mc = p.get("monotone_constraints")
+        if isinstance(mc, str):
+            body = mc.strip().strip("()[]")
+            mc = tuple(int(float(t)) for t in body.split(",") if t.strip()) if body else None
+        self.monotone_constraints = mc
+        ic = p.get("interaction_constraints")
+        if isinstance(ic, str):
+            import ast
+
+            ic = ast.literal_eval(ic) if ic.strip() else None
+        self.interaction_constraints = ic
         # Accepted for xgboost parity; this framework is ALWAYS deterministic.
         # A float-atomic LDS fast path for "false" was built and measured
         # 3x SLOWER than the int64 fixed-point slab (80.9 vs 239.4 r/s on the

@@ -94,3 +94,35 @@ def test_interaction_constraints_respected():
             walk(int(tree.right[nid]), feats)
 
         walk(0, set())
+
+
+def test_constraint_string_forms_accepted():
+    """The SageMaker hyperparameters arrive as strings — '(1,-1,0)' and
+    '[[0,1],[2,3]]' must parse exactly as xgboost does (the reference
+    forwards them verbatim to xgb.train)."""
+    import numpy as np
+
+    from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+    from sagemaker_xgboost_container_amd.models import trainer
+
+    rng = np.random.default_rng(5)
+    X = rng.normal(size=(3000, 4)).astype(np.float32)
+    y = (X[:, 0] - X[:, 1] + 0.2 * rng.normal(size=3000)).astype(np.float32)
+    dm = DMatrix(X, label=y)
+    base = {"objective": "reg:squarederror", "max_depth": 4, "device": "cpu"}
+
+    b_str = trainer.train(dict(base, monotone_constraints="(1,-1,0,0)"), dm,
+                          num_boost_round=4, verbose_eval=False)
+    b_tup = trainer.train(dict(base, monotone_constraints=(1, -1, 0, 0)), dm,
+                          num_boost_round=4, verbose_eval=False)
+    import json
+
+    assert json.dumps(b_str.save_json(), sort_keys=True) == json.dumps(
+        b_tup.save_json(), sort_keys=True
+    )
+
+    b_ic = trainer.train(dict(base, interaction_constraints="[[0,1],[2,3]]"), dm,
+                         num_boost_round=2, verbose_eval=False)
+    for tree in b_ic.trees:
+        feats_used = {int(f) for f, l in zip(tree.feature, tree.left) if l >= 0}
+        assert feats_used <= {0, 1} or feats_used <= {2, 3}
