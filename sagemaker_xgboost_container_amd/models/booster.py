@@ -131,14 +131,22 @@ class Booster:
         self._predict_cache = None
 
     # -- prediction ----------------------------------------------------------
-    def _margin(self, X, iteration_range=None):
-        """Raw margin (n,) or (n, k) for dense float32 tensor X."""
+    def _margin(self, X, iteration_range=None, base_margin=None):
+        """Raw margin (n,) or (n, k) for dense float32 tensor X.
+
+        `base_margin` (per-row, from the DMatrix) replaces the global
+        base_score as the starting margin — xgboost semantics."""
         obj = self.objective()
         n = X.shape[0]
         k = self.n_outputs
-        base = obj.base_margin(self.base_score)
         device = X.device
-        margin = torch.full((n, k), float(base), dtype=torch.float32, device=device)
+        if base_margin is not None:
+            margin = torch.as_tensor(
+                base_margin, dtype=torch.float32, device=device
+            ).reshape(n, k).clone()
+        else:
+            base = obj.base_margin(self.base_score)
+            margin = torch.full((n, k), float(base), dtype=torch.float32, device=device)
         lo, hi = 0, self.num_boosted_rounds()
         if iteration_range is not None and iteration_range != (0, 0):
             lo, hi = iteration_range
@@ -175,13 +183,14 @@ class Booster:
     ):
         """Predict for a DMatrix / ndarray. Returns numpy array."""
         X = self._as_tensor(data, validate_features)
+        base_margin = getattr(data, "get_base_margin", lambda: None)()
         if ntree_limit:  # legacy alias: trees -> iterations
             iteration_range = (0, int(ntree_limit) // max(1, self._trees_per_round()))
         if pred_contribs:
             return self._pred_contribs(X, approx=approx_contribs)
         if pred_leaf:
             return self._pred_leaf(X)
-        margin = self._margin(X, iteration_range)
+        margin = self._margin(X, iteration_range, base_margin=base_margin)
         if output_margin:
             return margin.cpu().numpy()
         out = self.objective().transform(margin)
