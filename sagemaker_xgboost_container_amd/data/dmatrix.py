@@ -360,13 +360,16 @@ class DMatrix:
         return sp.csr_matrix(np.nan_to_num(self._dense, nan=0.0))
 
     def slice(self, rindex):
-        """Row-subset DMatrix (used by k-fold CV, train.py:409-451 parity)."""
+        """Row-subset DMatrix (used by k-fold CV, train.py:409-451 parity).
+        Carries label/weight/base_margin; per-query groups are dropped
+        (row subsets break group alignment — xgboost does the same)."""
         rindex = np.asarray(rindex, dtype=np.int64)
         data = self._dense[rindex] if self._dense is not None else self._csr[rindex]
         out = DMatrix(
             data,
             label=self._label[rindex] if self._label is not None else None,
             weight=self._weight[rindex] if self._weight is not None else None,
+            base_margin=self._base_margin[rindex] if self._base_margin is not None else None,
             feature_names=self.feature_names,
         )
         return out
