@@ -210,6 +210,15 @@ class Booster:
                     f"feature_names mismatch: model expects {self.num_features} features, "
                     f"got {data.num_col()}"
                 )
+            if (
+                validate_features
+                and self.feature_names
+                and data.feature_names
+                and list(data.feature_names) != list(self.feature_names)
+            ):
+                raise ValueError(
+                    f"feature_names mismatch: {self.feature_names} vs {data.feature_names}"
+                )
             arr = data.to_dense()
         else:
             arr = np.asarray(data, dtype=np.float32)
