@@ -294,8 +294,40 @@ class Booster:
             return {name(f): covers[f] / counts[f] for f in counts}
         raise ValueError(f"Unknown importance type: {importance_type}")
 
+    def _tree_dump_json(self, tree, with_stats):
+        def name(f):
+            if self.feature_names and f < len(self.feature_names):
+                return self.feature_names[f]
+            return f"f{f}"
+
+        def node(nid, depth):
+            if tree.left[nid] < 0:
+                out = {"nodeid": int(nid), "leaf": float(tree.value[nid])}
+                if with_stats:
+                    out["cover"] = float(tree.sum_hess[nid])
+                return out
+            yes, no = int(tree.left[nid]), int(tree.right[nid])
+            out = {
+                "nodeid": int(nid),
+                "depth": depth,
+                "split": name(int(tree.feature[nid])),
+                "split_condition": float(tree.threshold[nid]),
+                "yes": yes,
+                "no": no,
+                "missing": yes if tree.default_left[nid] else no,
+            }
+            if with_stats:
+                out["gain"] = float(tree.gain[nid])
+                out["cover"] = float(tree.sum_hess[nid])
+            out["children"] = [node(yes, depth + 1), node(no, depth + 1)]
+            return out
+
+        return json.dumps(node(0, 0))
+
     def get_dump(self, fmap="", with_stats=False, dump_format="text"):
-        """Per-tree text dumps in the xgboost format."""
+        """Per-tree dumps in the xgboost text or json format."""
+        if dump_format == "json":
+            return [self._tree_dump_json(t, with_stats) for t in self.trees]
         dumps = []
         for tree in self.trees:
             lines = []

@@ -42,3 +42,31 @@ def test_pred_leaf():
     for t in range(4):
         tree = bst.trees[t]
         assert all(tree.left[nid] < 0 for nid in leaves[:, t])
+
+
+def test_get_dump_json_format():
+    import json as _json
+
+    import numpy as np
+
+    from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+    from sagemaker_xgboost_container_amd.models import trainer
+
+    rng = np.random.default_rng(0)
+    X = rng.normal(size=(500, 4)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(np.float32)
+    bst = trainer.train({"objective": "binary:logistic", "max_depth": 3, "device": "cpu"},
+                        DMatrix(X, label=y), num_boost_round=2, verbose_eval=False)
+    dumps = bst.get_dump(dump_format="json", with_stats=True)
+    assert len(dumps) == 2
+    root = _json.loads(dumps[0])
+    assert root["nodeid"] == 0
+    assert "split" in root and "children" in root and "gain" in root
+    # walk: every internal node has yes/no/missing; leaves carry values
+    def walk(n):
+        if "leaf" in n:
+            assert "cover" in n
+            return 1
+        assert n["missing"] in (n["yes"], n["no"])
+        return 1 + sum(walk(c) for c in n["children"])
+    assert walk(root) >= 3
