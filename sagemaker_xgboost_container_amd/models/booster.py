@@ -56,7 +56,8 @@ def objective_params_from_json(obj_json, params):
 
 
 class Booster:
-    def __init__(self, params=None, num_features=0, feature_names=None):
+    def __init__(self, params=None, num_features=0, feature_names=None,
+                 cache=None, model_file=None):
         self.params = dict(params or {})
         self.trees = []           # flat list of Tree
         self.tree_info = []       # class id of each tree (0 for single-output)
@@ -70,6 +71,8 @@ class Booster:
         self.best_score = None
         self._objective = None
         self._predict_cache = None
+        if model_file is not None:  # xgb.Booster(model_file=...) parity
+            self.load_model(model_file)
 
     # -- basic accessors ---------------------------------------------------
     @property
@@ -323,6 +326,36 @@ class Booster:
             return out
 
         return json.dumps(node(0, 0))
+
+    def trees_to_dataframe(self, fmap=""):
+        """Forest as a pandas DataFrame (xgboost column layout)."""
+        import pandas as pd
+
+        rows = []
+        for t, tree in enumerate(self.trees):
+            for nid in range(tree.num_nodes):
+                leaf = tree.left[nid] < 0
+                f = int(tree.feature[nid])
+                fname = (
+                    self.feature_names[f]
+                    if self.feature_names and f < len(self.feature_names)
+                    else f"f{f}"
+                )
+                rows.append({
+                    "Tree": t,
+                    "Node": int(nid),
+                    "ID": f"{t}-{nid}",
+                    "Feature": "Leaf" if leaf else fname,
+                    "Split": None if leaf else float(tree.threshold[nid]),
+                    "Yes": None if leaf else f"{t}-{int(tree.left[nid])}",
+                    "No": None if leaf else f"{t}-{int(tree.right[nid])}",
+                    "Missing": None if leaf else (
+                        f"{t}-{int(tree.left[nid]) if tree.default_left[nid] else int(tree.right[nid])}"
+                    ),
+                    "Gain": float(tree.value[nid]) if leaf else float(tree.gain[nid]),
+                    "Cover": float(tree.sum_hess[nid]),
+                })
+        return pd.DataFrame(rows)
 
     def get_dump(self, fmap="", with_stats=False, dump_format="text"):
         """Per-tree dumps in the xgboost text or json format."""
