@@ -219,6 +219,8 @@ class DMatrix:
         feature_names=None,
         nthread=None,
         base_margin=None,
+        qid=None,
+        group=None,
     ):
         self.missing = np.nan if missing is None else missing
         self.feature_names = feature_names
@@ -264,6 +266,11 @@ class DMatrix:
         self._label = None if label is None else np.asarray(label, dtype=np.float32).reshape(-1)
         self._weight = None if weight is None else np.asarray(weight, dtype=np.float32).reshape(-1)
         self._base_margin = None if base_margin is None else np.asarray(base_margin, dtype=np.float32)
+
+        if qid is not None:
+            self.set_qid(np.asarray(qid))
+        if group is not None:
+            self.set_group(group)
 
         if self._label is not None and len(self._label) != self.num_row():
             raise exc.UserError(

@@ -147,10 +147,13 @@ def train(
     verbose_eval=True,
     xgb_model=None,
     callbacks=None,
+    custom_metric=None,
     comm=None,
 ):
     """Train a Booster. API-parity with xgboost.train (reference call sites
     train.py:367-376,432-442; checkpointing.py:74)."""
+    if feval is None and custom_metric is not None:
+        feval = custom_metric  # xgboost >= 1.6 alias
     params = dict(params or {})
     unknown = set(params) - _TRAIN_PARAM_KEYS
     for key in unknown:

@@ -18,6 +18,33 @@ from .sklearn import XGBClassifier, XGBModel, XGBRanker, XGBRegressor  # noqa: F
 
 __version__ = "3.0.5"
 
+_GLOBAL_CONFIG = {"verbosity": 1, "use_rmm": False}
+
+
+def set_config(**kwargs):
+    _GLOBAL_CONFIG.update(kwargs)
+
+
+def get_config():
+    return dict(_GLOBAL_CONFIG)
+
+
+class config_context:
+    def __init__(self, **kwargs):
+        self._new = kwargs
+        self._old = None
+
+    def __enter__(self):
+        self._old = get_config()
+        set_config(**self._new)
+        return self
+
+    def __exit__(self, *exc):
+        _GLOBAL_CONFIG.clear()
+        _GLOBAL_CONFIG.update(self._old)
+        return False
+
+
 
 def plot_importance(booster, **kwargs):
     """Matplotlib feature-importance bar plot (xgboost.plot_importance)."""
