@@ -111,3 +111,77 @@ class TestUbjsonProperties:
             return x
 
         assert norm(decoded) == norm(json.loads(json.dumps(doc)))
+
+
+@settings(max_examples=12, deadline=None)
+@given(
+    n=st.integers(min_value=30, max_value=400),
+    f=st.integers(min_value=2, max_value=10),
+    depth=st.integers(min_value=1, max_value=5),
+    nan_frac=st.floats(min_value=0.0, max_value=0.4),
+    seed=st.integers(min_value=0, max_value=10_000),
+)
+def test_treeshap_additivity_property(n, f, depth, nan_frac, seed):
+    """Exact TreeSHAP contributions sum to the margin for ANY tree/data
+    shape (incl. NaN patterns) — the defining invariant."""
+    import numpy as np
+
+    from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+    from sagemaker_xgboost_container_amd.models import trainer
+
+    rng = np.random.default_rng(seed)
+    X = rng.normal(size=(n, f)).astype(np.float32)
+    if nan_frac:
+        X[rng.random(size=X.shape) < nan_frac] = np.nan
+    y = (np.nan_to_num(X[:, 0]) > 0).astype(np.float32)
+    bst = trainer.train(
+        {"objective": "binary:logistic", "max_depth": depth, "device": "cpu"},
+        DMatrix(X, label=y), num_boost_round=2, verbose_eval=False,
+    )
+    contribs = bst.predict(X[: min(n, 40)], pred_contribs=True)
+    margin = bst.predict(X[: min(n, 40)], output_margin=True)
+    np.testing.assert_allclose(contribs.sum(axis=1), margin, rtol=1e-3, atol=1e-3)
+
+
+@settings(max_examples=10, deadline=None)
+@given(
+    n=st.integers(min_value=80, max_value=400),
+    f=st.integers(min_value=64, max_value=150),
+    density=st.floats(min_value=0.02, max_value=0.4),
+    seed=st.integers(min_value=0, max_value=10_000),
+)
+def test_sparse_dense_parity_property(n, f, density, seed):
+    """The CSR training backend grows bit-identical trees to the dense
+    path for ANY shape/density (incl. rows with zero features)."""
+    import json
+    import os
+
+    import numpy as np
+    import scipy.sparse as sp
+
+    from sagemaker_xgboost_container_amd.data.dmatrix import DMatrix
+    from sagemaker_xgboost_container_amd.models import trainer
+
+    rng = np.random.default_rng(seed)
+    nnz = max(1, int(n * f * density))
+    rows = rng.integers(0, n, nnz)
+    cols = rng.integers(0, f, nnz)
+    vals = rng.normal(size=nnz).astype(np.float32)
+    csr = sp.csr_matrix((vals, (rows, cols)), shape=(n, f), dtype=np.float32)
+    csr.sum_duplicates()
+    y = (rng.random(n) > 0.5).astype(np.float32)
+
+    sigs = {}
+    for mode in ("1", "0"):
+        os.environ["SMXGB_SPARSE"] = mode
+        try:
+            bst = trainer.train(
+                {"objective": "binary:logistic", "max_depth": 3, "device": "cpu"},
+                DMatrix(csr.copy(), label=y), num_boost_round=2, verbose_eval=False,
+            )
+        finally:
+            os.environ.pop("SMXGB_SPARSE", None)
+        sigs[mode] = json.dumps(
+            bst.save_json()["learner"]["gradient_booster"]["model"]["trees"], sort_keys=True
+        )
+    assert sigs["1"] == sigs["0"]
