@@ -66,6 +66,7 @@ class Booster:
         self.iteration_indptr = [0]
         self.num_features = num_features
         self.feature_names = feature_names
+        self.feature_types = None
         self.attributes_map = {}
         self.best_iteration = None
         self.best_score = None
@@ -356,6 +357,10 @@ class Booster:
                     "Cover": float(tree.sum_hess[nid]),
                 })
         return pd.DataFrame(rows)
+
+    def get_fscore(self, fmap=""):
+        """Alias of get_score(importance_type='weight') — xgboost parity."""
+        return self.get_score(fmap=fmap, importance_type="weight")
 
     def get_dump(self, fmap="", with_stats=False, dump_format="text"):
         """Per-tree dumps in the xgboost text or json format."""
