@@ -27,7 +27,22 @@ PARSED_MAX_CONTENT_LENGTH = int(os.getenv("MAX_CONTENT_LENGTH", "6291456"))
 
 
 def number_of_workers():
-    return int(os.getenv("SAGEMAKER_NUM_MODEL_WORKERS", multiprocessing.cpu_count()))
+    """Worker processes: reference semantics are cpu_count() gunicorn
+    workers (CPU predict, serve.py:38-39). With GPU predict each worker
+    owns a CUDA context on the same device and the 0.1 ms kernel saturates
+    long before cpu_count workers do — default to 4 there.
+    SAGEMAKER_NUM_MODEL_WORKERS always wins."""
+    env = os.getenv("SAGEMAKER_NUM_MODEL_WORKERS")
+    if env:
+        return int(env)
+    try:
+        import torch
+
+        if torch.cuda.is_available():
+            return min(4, multiprocessing.cpu_count())
+    except ImportError:  # pragma: no cover
+        pass
+    return multiprocessing.cpu_count()
 
 
 class ScoringService:
