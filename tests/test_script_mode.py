@@ -198,3 +198,24 @@ def test_reference_abalone_script_runs_unmodified(script_env, tmp_path, monkeypa
     bst = Booster()
     bst.load_model(str(model))
     assert len(bst.trees) == 5
+
+
+def test_sklearn_api_example_script(script_env, tmp_path):
+    """The shipped sklearn-API example (boston-pattern) runs in script
+    mode end to end: fit/predict, model save, xgb.cv artifact."""
+    repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    code_dir = tmp_path / "code_sk"
+    code_dir.mkdir()
+    src = open(os.path.join(repo_root, "examples", "sklearn_api_script_mode.py")).read()
+    (code_dir / "sk_script.py").write_text(src)
+    hp_file = script_env / "input/config/hyperparameters.json"
+    hp = {
+        "sagemaker_program": "sk_script.py",
+        "sagemaker_submit_directory": str(code_dir),
+        "n-estimators": "6",
+    }
+    hp_file.write_text(json.dumps(hp))
+    env = sm_env.TrainingEnv(base_path=str(script_env))
+    training.train(env)
+    assert (script_env / "model" / "xgboost-model").exists()
+    assert (script_env / "output/data/cv_results.csv").exists()
