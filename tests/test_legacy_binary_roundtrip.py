@@ -54,14 +54,20 @@ def write_legacy_binary(bst):
     out += head + bytes(136 - len(head))
     out += _pack_string(bst.objective_name)
     out += _pack_string(bst.booster_type)
-    gparam = struct.pack("<i", len(bst.trees))
-    out += gparam + bytes(160 - len(gparam))
-    for t in bst.trees:
-        out += _write_tree(t)
-    out += np.asarray(bst.tree_info, dtype="<i4").tobytes()
-    if bst.booster_type == "dart" and bst.trees:
-        wd = np.asarray(bst.weight_drop, dtype="<f4")
-        out += struct.pack("<Q", len(wd)) + wd.tobytes()
+    if bst.booster_type == "gblinear":
+        lparam = struct.pack("<Ii", bst.num_features, bst.n_outputs)
+        out += lparam + bytes(136 - len(lparam))
+        flat = np.asarray(bst.linear_model.to_flat(), dtype="<f4")
+        out += struct.pack("<Q", len(flat)) + flat.tobytes()
+    else:
+        gparam = struct.pack("<i", len(bst.trees))
+        out += gparam + bytes(160 - len(gparam))
+        for t in bst.trees:
+            out += _write_tree(t)
+        out += np.asarray(bst.tree_info, dtype="<i4").tobytes()
+        if bst.booster_type == "dart" and bst.trees:
+            wd = np.asarray(bst.weight_drop, dtype="<f4")
+            out += struct.pack("<Q", len(wd)) + wd.tobytes()
     if bst.attributes_map:
         out += struct.pack("<Q", len(bst.attributes_map))
         for k, v in bst.attributes_map.items():
@@ -94,6 +100,7 @@ def _train(objective, num_class=None, booster=None, rounds=3, seed=0, f=5):
         ("reg:squarederror", None, None),
         ("multi:softprob", 4, None),
         ("binary:logistic", None, "dart"),
+        ("reg:squarederror", None, "gblinear"),
     ],
 )
 def test_write_parse_roundtrip(objective, num_class, booster):
@@ -103,7 +110,8 @@ def test_write_parse_roundtrip(objective, num_class, booster):
     loaded = parse_legacy_binary(raw)
     assert loaded.objective_name == objective
     assert loaded.num_features == bst.num_features
-    assert len(loaded.trees) == len(bst.trees)
+    if booster != "gblinear":
+        assert len(loaded.trees) == len(bst.trees)
     assert loaded.attributes_map.get("note") == "roundtrip"
     np.testing.assert_allclose(loaded.predict(X[:40]), bst.predict(X[:40]), atol=1e-5)
 
